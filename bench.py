@@ -1,0 +1,174 @@
+#!/usr/bin/env python3
+"""Benchmark entry point (driver contract).
+
+    python bench.py --gpus N --steps K --warmup W
+
+Runs the flagship training step — multi-band (4-sub-band PQMF) MelGAN
+22.05 kHz bf16 training (BASELINE.json config 2; config 3 at N>1 via
+torchrun, one rank per GPU over RCCL) — on synthetic mels with
+random-init weights, and prints ONE JSON line on rank 0:
+
+    {"metric": "train_mel_frames_per_sec", "value": <aggregate over all
+     GPUs>, ...}
+
+Timing: W untimed warmup steps, then EXACTLY K steps bracketed by
+barrier + torch.cuda.synchronize() on both sides; elapsed is the MAX
+over ranks.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from melgan_amd.config import HParam, merge_default
+from melgan_amd.engine.trainer import Trainer
+from melgan_amd.parallel.dp import init_distributed
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--config", default=os.path.join(
+        os.path.dirname(os.path.abspath(__file__)), "configs",
+        "mb_22k_4band.yaml"))
+    ap.add_argument("--batch-size", type=int, default=None)
+    ap.add_argument("--adversarial", action="store_true", default=True,
+                    help="include discriminator phase (full GAN step)")
+    ap.add_argument("--no-adversarial", dest="adversarial",
+                    action="store_false")
+    ap.add_argument("--inference-rtf", action="store_true",
+                    help="also measure inference RTF @22.05 kHz")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        init_distributed()
+    have_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0"))) \
+        if have_gpu else torch.device("cpu")
+
+    hp = merge_default(HParam(args.config))
+    if args.batch_size:
+        hp.train.batch_size = args.batch_size
+    if not have_gpu:
+        hp.backend.amp = "none"
+        hp.backend.fused_adam = False
+    # bench measures the steady adversarial regime (full GAN step) unless
+    # --no-adversarial: warmup-phase-only steps would skip discriminator
+    # work inside the timed region
+    hp.train.discriminator_train_start_steps = 0 if args.adversarial \
+        else 10 ** 9
+    hp.log.log_dir = os.path.join("gpurun_out", "bench_logs") \
+        if os.path.isdir("gpurun_out") else "/tmp/melgan_bench_logs"
+
+    torch.manual_seed(1234 + rank)
+    trainer = Trainer(hp, f"bench_r{rank}", rank=rank, world_size=world,
+                      device=device, enable_tb=False)
+
+    B = hp.train.batch_size
+    seg = hp.audio.segment_length
+    frames = seg // hp.audio.hop_length
+    # pre-generate a small rotation of synthetic batches on device
+    gen = torch.Generator().manual_seed(4321 + rank)
+    batches = []
+    for _ in range(4):
+        mel = (torch.randn(B, hp.audio.n_mel_channels, frames,
+                           generator=gen) * 2.0 - 6.0).to(device)
+        audio = (torch.rand(B, 1, seg, generator=gen) * 1.6 - 0.8).to(device)
+        batches.append((mel, audio))
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        trainer.train_step(*batches[i % len(batches)])
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        trainer.train_step(*batches[i % len(batches)])
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if have_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t)
+
+    rtf = None
+    if args.inference_rtf and rank == 0:
+        rtf = measure_rtf(hp, device)
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        frames_per_step = B * frames * world
+        value = frames_per_step / (elapsed / args.steps)
+        out = {
+            "metric": "train_mel_frames_per_sec",
+            "value": round(value, 1),
+            "unit": "mel-frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if have_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "mb-melgan-4band-22k",
+                "global_batch": B * world,
+                "seq_len": frames,
+                "parallelism": f"dp{world}",
+                "segment_samples": seg,
+                "adversarial": args.adversarial,
+            },
+        }
+        if rtf is not None:
+            out["config"]["inference_rtf_22k"] = rtf
+        print(json.dumps(out))
+
+
+def measure_rtf(hp, device, n_frames: int = 860, reps: int = 20):
+    """Inference RTF @22.05 kHz: ~10 s of audio per run, fused fwd path."""
+    from melgan_amd.models import build_generator
+    from melgan_amd.ops.pqmf import PQMF
+
+    g = build_generator(hp).to(device).eval()
+    g.remove_weight_norm()
+    pqmf = PQMF(hp.model.out_channels).to(device) \
+        if hp.model.out_channels > 1 else None
+    mel = (torch.randn(1, hp.audio.n_mel_channels, n_frames,
+                       device=device) * 2.0 - 6.0)
+    sr = hp.audio.sampling_rate
+    with torch.no_grad():
+        for _ in range(3):
+            sub = g(mel)
+            audio = pqmf.synthesis(sub) if pqmf else sub
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            sub = g(mel)
+            audio = pqmf.synthesis(sub) if pqmf else sub
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / reps
+    audio_s = audio.size(-1) / sr
+    return round(dt / audio_s, 6)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,254 @@
+"""Training engine — the hot loop.
+
+Per step:
+    fake_sub = G(mel)                          # fused HIP convs
+    fake     = PQMF.synthesis(fake_sub)        # fused K9
+    stft losses (full on fake, sub-band on fake_sub vs PQMF.analysis(audio))
+    [after discriminator_train_start_steps] adversarial + feature matching
+    g backward  -> bucketed RCCL all-reduce overlapped (parallel.dp)
+    g step (fused multi-tensor Adam)
+    d loss/backward/step
+    periodic: log / validate / checkpoint
+
+bf16 autocast on the model forward; all loss math fp32.
+"""
+
+from __future__ import annotations
+
+import itertools
+import time
+
+import torch
+
+from melgan_amd.engine import losses as L
+from melgan_amd.engine.checkpoint import load_checkpoint, save_checkpoint
+from melgan_amd.engine.writer import MetricsWriter, get_logger
+from melgan_amd.models import build_discriminator, build_generator
+from melgan_amd.ops.pqmf import PQMF
+from melgan_amd.ops.stft_loss import MultiResolutionSTFTLoss
+from melgan_amd.parallel.dp import BucketAllReduceDP
+from melgan_amd.config import check_hp_compat
+
+
+def make_optimizers(hp, generator, discriminator, device):
+    from melgan_amd.ops.adam import FusedAdam
+
+    lr = hp.train.adam.lr
+    betas = (hp.train.adam.beta1, hp.train.adam.beta2)
+    use_fused = hp.backend.fused_adam and device.type == "cuda"
+    if use_fused:
+        og = FusedAdam(generator.parameters(), lr=lr, betas=betas)
+        od = FusedAdam(discriminator.parameters(), lr=lr, betas=betas)
+    else:
+        og = torch.optim.Adam(generator.parameters(), lr=lr, betas=betas)
+        od = torch.optim.Adam(discriminator.parameters(), lr=lr, betas=betas)
+    return og, od
+
+
+class Trainer:
+    def __init__(self, hp, name: str, rank: int = 0, world_size: int = 1,
+                 device: torch.device | None = None, enable_tb: bool = True):
+        self.hp = hp
+        self.name = name
+        self.rank = rank
+        self.world = world_size
+        if device is None:
+            device = torch.device("cuda", rank % max(
+                torch.cuda.device_count(), 1)) \
+                if torch.cuda.is_available() else torch.device("cpu")
+        self.device = device
+
+        self.generator = build_generator(hp).to(device)
+        self.discriminator = build_discriminator(hp).to(device)
+        self.multiband = hp.model.out_channels > 1
+        if self.multiband:
+            self.pqmf = PQMF(hp.model.out_channels,
+                             cutoff_ratio=hp.model.get("pqmf_cutoff")).to(device)
+        self.stft_full = MultiResolutionSTFTLoss(
+            resolutions=hp.train.stft_loss_params, subband=False).to(device)
+        if self.multiband and hp.model.use_subband_stft_loss:
+            self.stft_sub = MultiResolutionSTFTLoss(subband=True).to(device)
+        else:
+            self.stft_sub = None
+
+        self.optim_g, self.optim_d = make_optimizers(
+            hp, self.generator, self.discriminator, device)
+        self.step = 0
+
+        self.dp_g = BucketAllReduceDP(self.generator,
+                                      hp.backend.bucket_cap_mb)
+        self.dp_d = BucketAllReduceDP(self.discriminator,
+                                      hp.backend.bucket_cap_mb)
+
+        self.amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                          "none": None, "fp32": None}[hp.backend.amp or "none"]
+        self.use_amp = (self.amp_dtype is not None
+                        and self.device.type == "cuda")
+        self.logger = get_logger(hp.log.log_dir, name) if rank == 0 else None
+        self.writer = MetricsWriter(hp.log.log_dir, name,
+                                    enable_tb=enable_tb, rank=rank)
+        self.lambda_adv = hp.train.get("lambda_adv") or 2.5
+        self.lambda_feat = hp.model.feat_match
+
+    # ------------------------------------------------------------------
+    def load(self, chkpt_path: str):
+        ck = load_checkpoint(chkpt_path, map_location=self.device)
+        from melgan_amd.engine.checkpoint import load_flexible
+
+        load_flexible(self.generator, ck["model_g"], logger=self.logger)
+        load_flexible(self.discriminator, ck["model_d"], logger=self.logger)
+        try:
+            self.optim_g.load_state_dict(ck["optim_g"])
+            self.optim_d.load_state_dict(ck["optim_d"])
+        except Exception:
+            if self.logger:
+                self.logger.warning("optimizer state incompatible; "
+                                    "reinitialised")
+        self.step = ck.get("step", 0)
+        if "hp_str" in ck:
+            check_hp_compat(ck["hp_str"], self.hp.hp_str, self.logger)
+        if self.logger:
+            self.logger.info(f"resumed from {chkpt_path} at step {self.step}")
+
+    def save(self):
+        if self.rank != 0:
+            return None
+        return save_checkpoint(
+            self.hp.log.chkpt_dir, self.name, self.step, self.generator,
+            self.discriminator, self.optim_g, self.optim_d, self.hp.hp_str)
+
+    # ------------------------------------------------------------------
+    def _forward_g(self, mel, audio):
+        """Generator forward + all generator losses. Returns (loss, parts,
+        fake full-band audio)."""
+        hp = self.hp
+        with torch.autocast("cuda", dtype=self.amp_dtype,
+                            enabled=self.use_amp):
+            fake_sub = self.generator(mel)  # (B, N_bands, S/N)
+        fake_sub = fake_sub.float()
+        if self.multiband:
+            fake = self.pqmf.synthesis(fake_sub)  # (B, 1, S)
+        else:
+            fake = fake_sub
+
+        sc_full, mag_full = self.stft_full(fake.squeeze(1),
+                                           audio.squeeze(1).float())
+        stft_loss = sc_full + mag_full
+        sub_loss = None
+        if self.stft_sub is not None:
+            with torch.no_grad():
+                audio_sub = self.pqmf.analysis(audio.float())
+            sc_sub, mag_sub = self.stft_sub(fake_sub, audio_sub)
+            sub_loss = sc_sub + mag_sub
+            stft_loss = 0.5 * stft_loss + 0.5 * sub_loss
+
+        g_loss = stft_loss
+        adv = feat = None
+        if self.step >= hp.train.discriminator_train_start_steps:
+            with torch.autocast("cuda", dtype=self.amp_dtype,
+                                enabled=self.use_amp):
+                d_fake = self.discriminator(fake)
+                with torch.no_grad():
+                    d_real = self.discriminator(audio.float())
+            adv = L.adversarial_g_loss(d_fake)
+            feat = L.feature_matching_loss(d_fake, d_real)
+            g_loss = g_loss + self.lambda_adv * adv + self.lambda_feat * feat
+        return g_loss, dict(stft=stft_loss, sub=sub_loss, adv=adv,
+                            feat=feat), fake
+
+    def train_step(self, mel, audio):
+        hp = self.hp
+        mel = mel.to(self.device, non_blocking=True)
+        audio = audio.to(self.device, non_blocking=True)
+
+        # ---- generator ------------------------------------------------
+        self.optim_g.zero_grad(set_to_none=True)
+        g_loss, parts, fake = self._forward_g(mel, audio)
+        g_loss.backward()
+        self.dp_g.finish()
+        self.optim_g.step()
+
+        # ---- discriminator --------------------------------------------
+        d_loss = None
+        if self.step >= hp.train.discriminator_train_start_steps:
+            self.optim_d.zero_grad(set_to_none=True)
+            with torch.autocast("cuda", dtype=self.amp_dtype,
+                                enabled=self.use_amp):
+                d_real = self.discriminator(audio.float())
+                d_fake = self.discriminator(fake.detach())
+            d_loss = L.discriminator_loss(d_real, d_fake)
+            d_loss.backward()
+            self.dp_d.finish()
+            self.optim_d.step()
+
+        self.step += 1
+        return g_loss.detach(), parts, \
+            d_loss.detach() if d_loss is not None else None
+
+    # ------------------------------------------------------------------
+    def train(self, trainloader, valloader=None, max_steps: int | None = None):
+        hp = self.hp
+        max_steps = max_steps or hp.train.max_steps
+        frames_per_batch = (hp.train.batch_size *
+                            hp.audio.segment_length // hp.audio.hop_length)
+        t0 = time.time()
+        steps_t0 = self.step
+        data_iter = itertools.cycle(trainloader) \
+            if hasattr(trainloader, "__iter__") else trainloader
+        for mel, audio in data_iter:
+            if self.step >= max_steps:
+                break
+            g_loss, parts, d_loss = self.train_step(mel, audio)
+
+            if self.step % hp.log.summary_interval == 0 and self.rank == 0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                dt = time.time() - t0
+                nsteps = self.step - steps_t0
+                mfps = (nsteps * frames_per_batch * self.world / dt) \
+                    if dt > 0 else 0.0
+                self.writer.log_training(
+                    self.step, g_loss=g_loss, d_loss=d_loss,
+                    stft=parts["stft"], sub=parts["sub"], adv=parts["adv"],
+                    feat=parts["feat"], mel_frames_per_sec=mfps)
+                if self.logger:
+                    self.logger.info(
+                        f"step {self.step} g_loss {float(g_loss):.4f}"
+                        + (f" d_loss {float(d_loss):.4f}" if d_loss is not None
+                           else "")
+                        + f" | {mfps:,.0f} mel-frames/s")
+                t0 = time.time()
+                steps_t0 = self.step
+
+            if valloader is not None and \
+                    self.step % hp.log.validation_interval == 0:
+                self.validate(valloader)
+
+            if self.step % hp.log.save_interval == 0:
+                self.save()
+        return self.step
+
+    @torch.no_grad()
+    def validate(self, valloader, max_batches: int = 8):
+        self.generator.eval()
+        tot_sc = tot_mag = 0.0
+        n = 0
+        last_audio = None
+        for i, (mel, audio) in enumerate(valloader):
+            if i >= max_batches:
+                break
+            mel = mel.to(self.device)
+            audio = audio.to(self.device)
+            fake_sub = self.generator(mel).float()
+            fake = self.pqmf.synthesis(fake_sub) if self.multiband else fake_sub
+            sc, mag = self.stft_full(fake.squeeze(1), audio.squeeze(1).float())
+            tot_sc += float(sc)
+            tot_mag += float(mag)
+            n += 1
+            last_audio = fake[0, 0].float().cpu()
+        self.generator.train()
+        if n and self.rank == 0:
+            self.writer.log_validation(
+                self.step, audio=last_audio, sr=self.hp.audio.sampling_rate,
+                val_sc=tot_sc / n, val_mag=tot_mag / n)
+        return (tot_sc / n, tot_mag / n) if n else (0.0, 0.0)

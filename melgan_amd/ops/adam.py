@@ -1,0 +1,73 @@
+"""Multi-tensor fused Adam.
+
+One HIP kernel launch updates ALL parameter tensors of a model per step
+(the generator has ~80 small conv weights; per-tensor eager Adam is
+launch-bound).  Tensors are described by a device-side pointer table;
+the kernel grid-strides over the concatenated element range.
+
+Falls back to torch.optim.Adam semantics exactly (same state dict
+layout: step / exp_avg / exp_avg_sq per param) so checkpoints are
+interchangeable with the reference's optimizer state.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from melgan_amd.ops import _ext
+
+
+class FusedAdam(torch.optim.Optimizer):
+    def __init__(self, params, lr=1e-4, betas=(0.5, 0.9), eps=1e-8,
+                 weight_decay=0.0):
+        defaults = dict(lr=lr, betas=betas, eps=eps,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            params, grads, exp_avgs, exp_avg_sqs, steps = [], [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = torch.zeros((), dtype=torch.float32)
+                    state["exp_avg"] = torch.zeros_like(p)
+                    state["exp_avg_sq"] = torch.zeros_like(p)
+                params.append(p)
+                grads.append(p.grad)
+                exp_avgs.append(state["exp_avg"])
+                exp_avg_sqs.append(state["exp_avg_sq"])
+                state["step"] += 1
+                steps.append(int(state["step"]))
+            if not params:
+                continue
+            beta1, beta2 = group["betas"]
+            e = _ext.ext()
+            if e is not None and params[0].is_cuda and hasattr(
+                    e, "multi_tensor_adam"):
+                e.multi_tensor_adam(
+                    params, grads, exp_avgs, exp_avg_sqs,
+                    group["lr"], beta1, beta2, group["eps"],
+                    group["weight_decay"], steps[0])
+            else:
+                # torch's own multi-tensor path (same math)
+                torch._foreach_mul_(exp_avgs, beta1)
+                torch._foreach_add_(exp_avgs, grads, alpha=1 - beta1)
+                torch._foreach_mul_(exp_avg_sqs, beta2)
+                torch._foreach_addcmul_(exp_avg_sqs, grads, grads,
+                                        value=1 - beta2)
+                for p, m, v, t in zip(params, exp_avgs, exp_avg_sqs, steps):
+                    bc1 = 1 - beta1 ** t
+                    bc2 = 1 - beta2 ** t
+                    denom = (v / bc2).sqrt_().add_(group["eps"])
+                    if group["weight_decay"]:
+                        p.mul_(1 - group["lr"] * group["weight_decay"])
+                    p.addcdiv_(m, denom, value=-group["lr"] / bc1)
+        return loss

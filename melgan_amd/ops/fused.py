@@ -1,0 +1,268 @@
+"""Autograd Functions over the gfx950 HIP kernels (melgan_amd._C).
+
+Kernel API contract (implemented in melgan_amd/kernels/):
+
+  conv1d_fwd(x, w, bias?, stride, dilation, pad, reflect, groups,
+             act, slope, act_before) -> y
+      direct conv; reflection/zero pad folded into addressing; LeakyReLU
+      prologue (act_before) or LeakyReLU/Tanh epilogue fused.
+  conv1d_bwd_data(gz, w, stride, dilation, pad, reflect, groups, T_in) -> gx
+  conv1d_bwd_weight(gz, x, stride, dilation, pad, reflect, groups, K,
+                    act_before, slope) -> gw (fp32)
+  conv1d_bias_grad(gz) -> gbias (fp32)
+  gather_conv_fwd(x, w, bias?, stride_up, pad, act_slope, act_before,
+                  tanh_out) -> y      # transposed conv, output-stationary
+  gather_conv_bwd_data / gather_conv_bwd_weight
+  act_bwd(gy, y, act, slope) -> gz    # epilogue activation backward
+  leaky_bwd_inplace(gx, x, slope)     # prologue activation backward
+  multi_tensor_adam(...)
+
+Weight-norm chain rule (w = g*v/||v||) is applied HERE in fp32 (the
+per-channel reductions are tiny next to the convs — SURVEY.md §2.2 K5
+is eliminated from the hot path by folding w into the conv kernels).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from melgan_amd.ops import _ext
+from melgan_amd.ops.wnconv import ACT_LEAKY, ACT_NONE, ACT_TANH
+
+
+def _wn_weight_and_norm(v: torch.Tensor, g: torch.Tensor):
+    n = v.flatten(1).norm(dim=1).clamp_min(1e-12)  # (C0,)
+    shape = (-1,) + (1,) * (v.dim() - 1)
+    w = v * (g.flatten() / n).reshape(shape)
+    return w, n
+
+
+def _wn_backward(gw: torch.Tensor, v: torch.Tensor, g: torch.Tensor,
+                 n: torch.Tensor):
+    """grads of w = g*v/n wrt v and g; all fp32."""
+    gw = gw.float()
+    v = v.float()
+    gflat = g.flatten().float()
+    shape = (-1,) + (1,) * (v.dim() - 1)
+    dot = (gw * v).flatten(1).sum(1)  # (C0,)
+    grad_g = (dot / n).reshape(g.shape)
+    grad_v = (gflat / n).reshape(shape) * gw \
+        - ((gflat * dot) / (n ** 3)).reshape(shape) * v
+    return grad_v, grad_g
+
+
+class FusedWNConv1dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, v, g, bias, stride, dilation, pad, reflect, groups,
+                act, slope, act_before):
+        C = _ext.require_ext()
+        w, n = _wn_weight_and_norm(v.float(), g.float())
+        w_c = w.to(x.dtype)
+        b_c = bias.to(x.dtype) if bias is not None else None
+        y = C.conv1d_fwd(x, w_c, b_c, stride, dilation, pad, reflect,
+                         groups, act, slope, act_before)
+        ctx.save_for_backward(x, v, g, n, w_c, y)
+        ctx.geom = (stride, dilation, pad, reflect, groups, act, slope,
+                    act_before)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        x, v, g, n, w_c, y = ctx.saved_tensors
+        stride, dilation, pad, reflect, groups, act, slope, act_before = \
+            ctx.geom
+        gy = gy.contiguous()
+        if act != ACT_NONE and not act_before:
+            gz = C.act_bwd(gy, y, act, slope)
+        else:
+            gz = gy
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = C.conv1d_bwd_data(gz, w_c, stride, dilation, pad, reflect,
+                                   groups, x.size(2))
+            if act_before:
+                C.leaky_bwd_inplace(gx, x, slope)
+        gw_full = C.conv1d_bwd_weight(gz, x, stride, dilation, pad, reflect,
+                                      groups, w_c.size(2), act_before, slope)
+        gv, gg = _wn_backward(gw_full, v, g, n)
+        if ctx.has_bias:
+            gb = C.conv1d_bias_grad(gz)
+        return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
+                None, None, None, None, None)
+
+
+def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
+    """Module-level entry: dispatch WNConv1d to the fused kernel."""
+    if torch.is_autocast_enabled("cuda") and x.dtype == torch.float32:
+        x = x.to(torch.get_autocast_dtype("cuda"))
+    x = x.contiguous()
+    if mod._removed:
+        C = _ext.require_ext()
+        w_c = mod.weight.to(x.dtype)
+        b_c = mod.bias.to(x.dtype) if mod.bias is not None else None
+        return C.conv1d_fwd(x, w_c, b_c, mod.stride, mod.dilation, mod.pad,
+                            mod.pad_mode == "reflect", mod.groups, mod.act,
+                            mod.act_slope, mod.act_before)
+    return FusedWNConv1dFn.apply(
+        x, mod.weight_v, mod.weight_g, mod.bias, mod.stride, mod.dilation,
+        mod.pad, mod.pad_mode == "reflect", mod.groups, mod.act,
+        mod.act_slope, mod.act_before)
+
+
+class FusedWNConvT1dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, v, g, bias, stride, pad, out_pad, slope, act_before):
+        C = _ext.require_ext()
+        w, n = _wn_weight_and_norm(v.float(), g.float())  # (Cin, Cout, K)
+        w_c = w.to(x.dtype)
+        b_c = bias.to(x.dtype) if bias is not None else None
+        y = C.conv_transpose1d_fwd(x, w_c, b_c, stride, pad, out_pad,
+                                   slope, act_before)
+        ctx.save_for_backward(x, v, g, n, w_c)
+        ctx.geom = (stride, pad, out_pad, slope, act_before)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        x, v, g, n, w_c = ctx.saved_tensors
+        stride, pad, out_pad, slope, act_before = ctx.geom
+        gy = gy.contiguous()
+        gx = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = C.conv_transpose1d_bwd_data(gy, w_c, stride, pad, out_pad,
+                                             x.size(2))
+            if act_before:
+                C.leaky_bwd_inplace(gx, x, slope)
+        gw_full = C.conv_transpose1d_bwd_weight(gy, x, stride, pad, out_pad,
+                                                w_c.size(2), act_before,
+                                                slope)
+        gv, gg = _wn_backward(gw_full, v, g, n)
+        if ctx.has_bias:
+            gb = C.conv1d_bias_grad(gy)
+        return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
+                None, None)
+
+
+def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
+    if torch.is_autocast_enabled("cuda") and x.dtype == torch.float32:
+        x = x.to(torch.get_autocast_dtype("cuda"))
+    x = x.contiguous()
+    if mod._removed:
+        C = _ext.require_ext()
+        w_c = mod.weight.to(x.dtype)
+        b_c = mod.bias.to(x.dtype) if mod.bias is not None else None
+        return C.conv_transpose1d_fwd(x, w_c, b_c, mod.stride, mod.padding,
+                                      mod.output_padding, mod.act_slope,
+                                      mod.act_before)
+    return FusedWNConvT1dFn.apply(
+        x, mod.weight_v, mod.weight_g, mod.bias, mod.stride, mod.padding,
+        mod.output_padding, mod.act_slope, mod.act_before)
+
+
+# ---------------------------------------------------------------------------
+# PQMF — fixed filters, no weight grads
+# ---------------------------------------------------------------------------
+class PQMFAnalysisFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, filt, subbands, pad):
+        C = _ext.require_ext()
+        y = C.conv1d_fwd(x, filt, None, subbands, 1, pad, False, 1,
+                         ACT_NONE, 0.0, False)
+        ctx.save_for_backward(filt)
+        ctx.geom = (subbands, pad, x.size(2))
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        (filt,) = ctx.saved_tensors
+        subbands, pad, t_in = ctx.geom
+        gx = C.conv1d_bwd_data(gy.contiguous(), filt, subbands, 1, pad,
+                               False, 1, t_in)
+        return gx, None, None, None
+
+
+def pqmf_analysis(mod, x: torch.Tensor) -> torch.Tensor:
+    filt = mod.analysis_filter.to(x.dtype)
+    return PQMFAnalysisFn.apply(x.contiguous(), filt, mod.subbands, mod.pad)
+
+
+class PQMFSynthesisFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, filt_t, subbands, pad):
+        # filt_t: (N, 1, K) = flip(G)*N — conv_transpose weight layout
+        C = _ext.require_ext()
+        y = C.conv_transpose1d_fwd(x, filt_t, None, subbands, pad,
+                                   subbands - 1, 0.0, False)
+        ctx.save_for_backward(filt_t)
+        ctx.geom = (subbands, pad, x.size(2))
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        (filt_t,) = ctx.saved_tensors
+        subbands, pad, t_in = ctx.geom
+        gx = C.conv_transpose1d_bwd_data(gy.contiguous(), filt_t, subbands,
+                                         pad, subbands - 1, t_in)
+        return gx, None, None, None
+
+
+def pqmf_synthesis(mod, x: torch.Tensor) -> torch.Tensor:
+    key = "_fused_synth_w"
+    w = getattr(mod, key, None)
+    if w is None or w.dtype != x.dtype or w.device != x.device:
+        # (1, N, K) conv weight -> (N, 1, K) transposed-conv weight, flipped
+        w = (mod.synthesis_filter * mod.subbands).flip(-1).to(x.dtype)
+        setattr(mod, key, w)
+    pad = mod.taps + 1 - 1 - mod.pad  # K-1-pad == pad for even taps
+    return PQMFSynthesisFn.apply(x.contiguous(), w, mod.subbands, pad)
+
+
+# ---------------------------------------------------------------------------
+# Fused STFT magnitude (frame+window prologue, |.| epilogue on rocFFT)
+# ---------------------------------------------------------------------------
+def fused_stft_mag(x, fft_size, hop_size, win_length, window):
+    C = _ext.ext()
+    if C is None or not hasattr(C, "stft_frame"):
+        # rocFFT via torch.stft — still fully on-GPU
+        xs = torch.stft(x.float(), fft_size, hop_size, win_length,
+                        window=window.float(), return_complex=True,
+                        center=True, pad_mode="reflect")
+        return torch.clamp(xs.abs().transpose(1, 2), min=1e-7)
+    return _FusedSTFTMagFn.apply(x, fft_size, hop_size, win_length, window)
+
+
+class _FusedSTFTMagFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, fft_size, hop_size, win_length, window):
+        C = _ext.require_ext()
+        xf = x.float().contiguous()
+        frames = C.stft_frame(xf, fft_size, hop_size, win_length,
+                              window.float())  # (B, n_frames, fft)
+        spec = torch.fft.rfft(frames, dim=-1)  # rocFFT batched R2C
+        mag = C.complex_mag_clamp(torch.view_as_real(spec), 1e-7)
+        ctx.save_for_backward(torch.view_as_real(spec), mag, window)
+        ctx.geom = (fft_size, hop_size, win_length, x.size(-1))
+        return mag
+
+    @staticmethod
+    def backward(ctx, gmag):
+        C = _ext.require_ext()
+        spec_ri, mag, window = ctx.saved_tensors
+        fft_size, hop_size, win_length, t_in = ctx.geom
+        # d|S| backward: gS = gmag * S / |S| (fp32)
+        gspec_ri = C.mag_bwd(gmag.contiguous(), spec_ri, mag)
+        gspec = torch.view_as_complex(gspec_ri)
+        # rfft vjp: double the doubled bins then irfft-style adjoint
+        n = fft_size
+        gspec = gspec.clone()
+        gspec[..., 1:(n + 1) // 2] *= 2.0
+        gframes = torch.fft.irfft(gspec.conj(), n=n, dim=-1) * n
+        gx = C.stft_frame_bwd(gframes.contiguous(), window.float(),
+                              hop_size, win_length, t_in)
+        return gx, None, None, None, None

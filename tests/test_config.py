@@ -1,0 +1,54 @@
+import os
+
+from melgan_amd.config import HParam, merge_default, _default_ratios
+
+
+def test_attribute_access(tiny_hp):
+    assert tiny_hp.audio.n_mel_channels == 80
+    assert tiny_hp.model.generator_ratio == [8, 8, 2, 2]
+    assert tiny_hp.train.adam.lr == 1e-4  # default filled
+    assert tiny_hp.model.nonexistent_key is None  # family Dotdict behaviour
+
+
+def test_hp_str_roundtrip(tiny_hp):
+    from melgan_amd.config import load_hparam_str
+
+    hp2 = merge_default(load_hparam_str(tiny_hp.hp_str))
+    assert hp2.audio.segment_length == tiny_hp.audio.segment_length
+    assert hp2.hp_str == tiny_hp.hp_str
+
+
+def test_yaml_configs_load():
+    cfg_dir = os.path.join(os.path.dirname(__file__), "..", "configs")
+    for name in os.listdir(cfg_dir):
+        hp = merge_default(HParam(os.path.join(cfg_dir, name)))
+        # total upsampling * out_channels must equal hop_length
+        total = hp.model.out_channels
+        for r in hp.model.generator_ratio:
+            total *= r
+        assert total == hp.audio.hop_length, name
+        assert hp.audio.segment_length % hp.audio.hop_length == 0, name
+
+
+def test_default_ratios():
+    assert _default_ratios(256) == [8, 8, 2, 2]
+    for total in (256, 64, 32, 75, 300):
+        r = _default_ratios(total)
+        prod = 1
+        for x in r:
+            prod *= x
+        assert prod == total, (total, r)
+
+
+def test_reference_config_without_backend_group():
+    """Reference configs (no backend: group) must load unchanged."""
+    raw = """
+data: {train_dir: /tmp/x}
+train: {batch_size: 4}
+audio: {hop_length: 256}
+model: {out_channels: 4}
+log: {chkpt_dir: c, log_dir: l}
+"""
+    hp = merge_default(HParam(raw=raw))
+    assert hp.backend.use_hip_kernels is True
+    assert hp.model.generator_ratio is not None

@@ -1,0 +1,86 @@
+"""Distributed DP engine tests without a GPU: the bucketed
+all-reduce engine over the gloo backend, world_size=2, on 127.0.0.1."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from melgan_amd.models.generator import Generator
+
+
+def _make_model(seed=0):
+    torch.manual_seed(seed)
+    return Generator(mel_channels=16, out_channels=2, base_channels=16,
+                     ratios=[4, 2], n_residual_layers=2)
+
+
+def _worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from melgan_amd.parallel.dp import BucketAllReduceDP
+
+        model = _make_model(seed=123 + rank)  # different init per rank:
+        # broadcast at engine construction must equalise them
+        engine = BucketAllReduceDP(model, bucket_cap_mb=0.05)  # tiny buckets
+        # -> multiple buckets exercise the launch-per-bucket path
+
+        torch.manual_seed(1000 + rank)  # different DATA per rank
+        mel = torch.randn(2, 16, 8)
+        y = model(mel)
+        y.pow(2).mean().backward()
+        engine.finish()
+
+        grads = torch.cat([p.grad.flatten() for p in model.parameters()])
+        params = torch.cat([p.data.flatten() for p in model.parameters()])
+        results[rank] = (grads, params)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_bucketed_allreduce_matches_large_batch():
+    world = 2
+    port = 29531
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker, args=(world, port, results), nprocs=world, join=True)
+
+    g0, p0 = results[0]
+    g1, p1 = results[1]
+    # both ranks ended with identical (averaged) gradients and identical params
+    assert torch.allclose(g0, g1, atol=1e-6)
+    assert torch.equal(p0, p1)
+
+    # single-process reference: same model (rank0 init), batch = both ranks'
+    model = _make_model(seed=123)  # engine broadcast rank0 init
+    grads_accum = None
+    for rank in range(world):
+        torch.manual_seed(1000 + rank)
+        mel = torch.randn(2, 16, 8)
+        model.zero_grad()
+        y = model(mel)
+        y.pow(2).mean().backward()
+        g = torch.cat([p.grad.flatten() for p in model.parameters()])
+        grads_accum = g if grads_accum is None else grads_accum + g
+    ref = grads_accum / world
+    assert torch.allclose(g0, ref, atol=1e-5), \
+        (g0 - ref).abs().max().item()
+
+
+@pytest.mark.timeout(60)
+def test_world1_noop():
+    from melgan_amd.parallel.dp import BucketAllReduceDP
+
+    model = _make_model()
+    engine = BucketAllReduceDP(model)
+    assert not engine.enabled
+    y = model(torch.randn(1, 16, 8))
+    y.mean().backward()
+    engine.finish()  # no-op, must not raise

@@ -1,0 +1,82 @@
+import pytest
+import torch
+
+from melgan_amd.models import build_generator
+from melgan_amd.models.generator import Generator
+
+
+def test_forward_shape_singleband(tiny_hp):
+    g = build_generator(tiny_hp)
+    mel = torch.randn(2, 80, 16)
+    audio = g(mel)
+    assert audio.shape == (2, 1, 16 * 256)
+    assert audio.abs().max() <= 1.0  # tanh output
+
+
+def test_forward_shape_multiband(tiny_mb_hp):
+    g = build_generator(tiny_mb_hp)
+    mel = torch.randn(2, 80, 16)
+    audio = g(mel)
+    # 4 sub-bands, each upsampled 64x
+    assert audio.shape == (2, 4, 16 * 64)
+
+
+@pytest.mark.parametrize("ratios,out_ch", [
+    ([8, 8, 2, 2], 1), ([8, 4, 2], 4), ([5, 5, 3], 4), ([8, 4, 2], 8)])
+def test_all_config_geometries(ratios, out_ch):
+    g = Generator(mel_channels=80, out_channels=out_ch, base_channels=32,
+                  ratios=ratios)
+    t = 8
+    up = 1
+    for r in ratios:
+        up *= r
+    y = g(torch.randn(1, 80, t))
+    assert y.shape == (1, out_ch, t * up)
+
+
+def test_remove_weight_norm_identical_output(tiny_hp):
+    g = build_generator(tiny_hp)
+    g.eval()
+    mel = torch.randn(1, 80, 12)
+    with torch.no_grad():
+        y1 = g(mel)
+        g.remove_weight_norm()
+        y2 = g(mel)
+    assert torch.allclose(y1, y2, atol=1e-5), \
+        (y1 - y2).abs().max().item()
+
+
+def test_weight_norm_param_naming(tiny_hp):
+    """Family checkpoint naming contract: *.weight_g / *.weight_v / *.bias"""
+    g = build_generator(tiny_hp)
+    keys = list(g.state_dict().keys())
+    assert any(k.endswith("weight_g") for k in keys)
+    assert any(k.endswith("weight_v") for k in keys)
+    assert any(k.endswith("bias") for k in keys)
+    # no plain .weight before remove_weight_norm
+    assert not any(k.endswith(".weight") for k in keys)
+
+
+def test_inference_helper(tiny_hp):
+    g = build_generator(tiny_hp)
+    g.eval()
+    mel = torch.randn(80, 20)
+    audio = g.inference(mel)
+    assert audio.shape == (1, 1, 20 * 256)
+
+
+def test_backward_flows(tiny_mb_hp):
+    g = build_generator(tiny_mb_hp)
+    mel = torch.randn(2, 80, 8)
+    y = g(mel)
+    y.pow(2).mean().backward()
+    grads = [p.grad for p in g.parameters()]
+    assert all(gr is not None for gr in grads)
+    assert all(torch.isfinite(gr).all() for gr in grads)
+
+
+def test_param_count_order():
+    """4-band generator should be in the MB-MelGAN ~2M param class."""
+    g = Generator(80, 4, 512, [8, 4, 2], 3)
+    n = sum(p.numel() for p in g.parameters())
+    assert 1e6 < n < 6e6, n

@@ -1,0 +1,63 @@
+import torch
+
+from melgan_amd.data import create_dataloader
+from melgan_amd.engine.trainer import Trainer
+
+
+def test_train_steps_singleband(tiny_hp, tmp_path):
+    """Config-1 class smoke: a few CPU steps, finite losses, ckpt+resume."""
+    tiny_hp.log.chkpt_dir = str(tmp_path / "ck")
+    tiny_hp.log.log_dir = str(tmp_path / "logs")
+    tiny_hp.log.save_interval = 3
+    tr = Trainer(tiny_hp, "smoke", enable_tb=False)
+    dl = create_dataloader(tiny_hp, train=True, synthetic=True)
+    end = tr.train(dl, valloader=None, max_steps=3)
+    assert end == 3
+    # checkpoint written at step 3
+    ck_files = list((tmp_path / "ck").glob("smoke_*.pt"))
+    assert ck_files, "no checkpoint written"
+
+    tr2 = Trainer(tiny_hp, "smoke2", enable_tb=False)
+    tr2.load(str(ck_files[0]))
+    assert tr2.step == 3
+
+
+def test_train_steps_multiband(tiny_mb_hp, tmp_path):
+    """Multi-band path: PQMF + sub-band STFT loss + discriminator phase."""
+    tiny_mb_hp.log.chkpt_dir = str(tmp_path / "ck")
+    tiny_mb_hp.log.log_dir = str(tmp_path / "logs")
+    tr = Trainer(tiny_mb_hp, "smoke_mb", enable_tb=False)
+    dl = create_dataloader(tiny_mb_hp, train=True, synthetic=True)
+
+    losses = []
+    it = iter(dl)
+    for _ in range(3):
+        mel, audio = next(it)
+        g_loss, parts, d_loss = tr.train_step(mel, audio)
+        losses.append(float(g_loss))
+        assert torch.isfinite(g_loss)
+    # discriminator phase kicked in at step >= 1
+    assert d_loss is not None and torch.isfinite(d_loss)
+    assert parts["adv"] is not None
+    assert parts["feat"] is not None
+    assert parts["sub"] is not None
+
+
+def test_validation(tiny_mb_hp, tmp_path):
+    tiny_mb_hp.log.log_dir = str(tmp_path / "logs")
+    tr = Trainer(tiny_mb_hp, "val_test", enable_tb=False)
+    dl = create_dataloader(tiny_mb_hp, train=False, synthetic=True)
+    sc, mag = tr.validate(dl, max_batches=1)
+    assert sc > 0 and mag > 0
+
+
+def test_generator_warmup_no_disc(tiny_mb_hp, tmp_path):
+    """Before discriminator_train_start_steps only STFT losses train."""
+    tiny_mb_hp.train.discriminator_train_start_steps = 1000
+    tiny_mb_hp.log.log_dir = str(tmp_path / "logs")
+    tr = Trainer(tiny_mb_hp, "warmup_test", enable_tb=False)
+    dl = create_dataloader(tiny_mb_hp, train=True, synthetic=True)
+    mel, audio = next(iter(dl))
+    g_loss, parts, d_loss = tr.train_step(mel, audio)
+    assert d_loss is None
+    assert parts["adv"] is None
