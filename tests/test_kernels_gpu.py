@@ -1,0 +1,295 @@
+"""GPU numerics tests: every gfx950 kernel vs the PyTorch fp32 composite
+reference on the same device (SURVEY.md §4 kernel unit tests).
+
+Shape matrix drawn from the 5 driver configs: generator convs (k=7 input,
+k=3 dil 1/3/9/27, k=1), upsample ratios 8/4/2/5/3, discriminator towers
+(k=15, k=41 s=4 grouped, k=5, k=3), PQMF 4/8-band.
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def _ext():
+    import melgan_amd._C as C
+
+    return C
+
+
+def _mkconv(cin, cout, k, **kw):
+    from melgan_amd.ops.wnconv import WNConv1d
+
+    torch.manual_seed(0)
+    return WNConv1d(cin, cout, k, **kw).to(DEV)
+
+
+def _cmp(a, b, rtol=1e-4, atol=1e-4, what=""):
+    a = a.float()
+    b = b.float()
+    err = (a - b).abs().max().item()
+    denom = b.abs().max().item() + 1e-8
+    assert torch.allclose(a, b, rtol=rtol, atol=atol), \
+        f"{what}: max abs err {err:.3e} (ref max {denom:.3e})"
+
+
+CONV_CASES = [
+    # (Cin, Cout, K, stride, dil, groups, pad_mode, T)
+    (80, 512, 7, 1, 1, 1, "reflect", 64),     # generator input conv
+    (256, 256, 3, 1, 1, 1, "reflect", 512),   # res stack d=1
+    (128, 128, 3, 1, 3, 1, "reflect", 777),   # res stack d=3, odd T
+    (64, 64, 3, 1, 9, 1, "reflect", 2048),    # res stack d=9
+    (96, 96, 3, 1, 27, 1, "reflect", 4096),   # deep-residual d=27
+    (64, 64, 1, 1, 1, 1, "none", 512),        # 1x1
+    (64, 4, 7, 1, 1, 1, "reflect", 4096),     # output conv (4-band)
+    (1, 16, 15, 1, 1, 1, "reflect", 4096),    # disc input
+    (16, 64, 41, 4, 1, 4, "zero", 4096),      # disc tower L1
+    (64, 256, 41, 4, 1, 16, "zero", 1024),    # disc tower L2
+    (256, 1024, 41, 4, 1, 64, "zero", 512),   # disc tower L3
+    (1024, 1024, 41, 4, 1, 256, "zero", 256),  # disc tower L4
+    (1024, 1024, 5, 1, 1, 1, "zero", 64),     # disc k5
+    (1024, 1, 3, 1, 1, 1, "zero", 64),        # disc score
+]
+
+
+@pytest.mark.parametrize("cin,cout,k,s,d,g,pm,t", CONV_CASES)
+def test_conv1d_fwd_fp32(cin, cout, k, s, d, g, pm, t):
+    mod = _mkconv(cin, cout, k, stride=s, dilation=d, groups=g, pad_mode=pm,
+                  act="leaky")
+    x = torch.randn(3, cin, t, device=DEV)
+    y_ref = mod._composite(x)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    y = fused_conv1d(mod, x)
+    _cmp(y, y_ref, what=f"conv fwd {cin}x{cout}k{k}s{s}d{d}g{g}{pm}")
+
+
+@pytest.mark.parametrize("cin,cout,k,s,d,g,pm,t", CONV_CASES[:10])
+def test_conv1d_backward_fp32(cin, cout, k, s, d, g, pm, t):
+    mod = _mkconv(cin, cout, k, stride=s, dilation=d, groups=g, pad_mode=pm,
+                  act="leaky")
+    torch.manual_seed(1)
+    x0 = torch.randn(2, cin, t, device=DEV)
+
+    # composite reference
+    x1 = x0.clone().requires_grad_(True)
+    y1 = mod._composite(x1)
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    ref_grads = {n: p.grad.clone() for n, p in mod.named_parameters()}
+    ref_gx = x1.grad.clone()
+    for p in mod.parameters():
+        p.grad = None
+
+    # fused path
+    from melgan_amd.ops.fused import fused_conv1d
+
+    x2 = x0.clone().requires_grad_(True)
+    y2 = fused_conv1d(mod, x2)
+    y2.backward(gy)
+    tag = f"conv bwd {cin}x{cout}k{k}s{s}d{d}g{g}{pm}"
+    _cmp(y2, y1, what=tag + " y")
+    _cmp(x2.grad, ref_gx, rtol=1e-3, atol=1e-3, what=tag + " gx")
+    for n, p in mod.named_parameters():
+        _cmp(p.grad, ref_grads[n], rtol=1e-3, atol=1e-3,
+             what=f"{tag} g{n}")
+
+
+def test_conv1d_act_before_prologue():
+    mod = _mkconv(32, 32, 3, pad_mode="reflect", act_before=True)
+    x0 = torch.randn(2, 32, 128, device=DEV)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    x1 = x0.clone().requires_grad_(True)
+    y1 = mod._composite(x1)
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    gx_ref = x1.grad.clone()
+    gv_ref = mod.weight_v.grad.clone()
+    for p in mod.parameters():
+        p.grad = None
+    x2 = x0.clone().requires_grad_(True)
+    y2 = fused_conv1d(mod, x2)
+    y2.backward(gy)
+    _cmp(y2, y1, what="act_before y")
+    _cmp(x2.grad, gx_ref, rtol=1e-3, atol=1e-3, what="act_before gx")
+    _cmp(mod.weight_v.grad, gv_ref, rtol=1e-3, atol=1e-3,
+         what="act_before gv")
+
+
+def test_conv1d_tanh_epilogue():
+    mod = _mkconv(16, 4, 7, pad_mode="reflect", act="tanh")
+    x0 = torch.randn(2, 16, 256, device=DEV)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    x1 = x0.clone().requires_grad_(True)
+    y1 = mod._composite(x1)
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    gx_ref = x1.grad.clone()
+    for p in mod.parameters():
+        p.grad = None
+    x2 = x0.clone().requires_grad_(True)
+    y2 = fused_conv1d(mod, x2)
+    y2.backward(gy)
+    _cmp(y2, y1, what="tanh y")
+    _cmp(x2.grad, gx_ref, rtol=1e-3, atol=1e-3, what="tanh gx")
+
+
+@pytest.mark.parametrize("ratio,cin,cout,t", [
+    (8, 512, 256, 64), (4, 256, 128, 512), (2, 128, 64, 2048),
+    (5, 64, 32, 80), (3, 32, 16, 400)])
+def test_conv_transpose_fwd_bwd(ratio, cin, cout, t):
+    from melgan_amd.ops.wnconv import WNConvTranspose1d
+    from melgan_amd.ops.fused import fused_conv_transpose1d
+
+    torch.manual_seed(0)
+    mod = WNConvTranspose1d(cin, cout, ratio).to(DEV)
+    x0 = torch.randn(2, cin, t, device=DEV)
+
+    x1 = x0.clone().requires_grad_(True)
+    y1 = mod._composite(x1)
+    assert y1.size(2) == t * ratio
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    gx_ref = x1.grad.clone()
+    refs = {n: p.grad.clone() for n, p in mod.named_parameters()}
+    for p in mod.parameters():
+        p.grad = None
+
+    x2 = x0.clone().requires_grad_(True)
+    y2 = fused_conv_transpose1d(mod, x2)
+    y2.backward(gy)
+    tag = f"convT r{ratio}"
+    _cmp(y2, y1, what=tag + " y")
+    _cmp(x2.grad, gx_ref, rtol=1e-3, atol=1e-3, what=tag + " gx")
+    for n, p in mod.named_parameters():
+        _cmp(p.grad, refs[n], rtol=1e-3, atol=1e-3, what=f"{tag} g{n}")
+
+
+@pytest.mark.parametrize("subbands", [4, 8])
+def test_pqmf_fused(subbands):
+    from melgan_amd.ops.pqmf import PQMF
+    from melgan_amd.ops import fused
+
+    torch.manual_seed(0)
+    pqmf = PQMF(subbands).to(DEV)
+    x = torch.randn(2, 1, 8192, device=DEV)
+    sub_ref = torch.nn.functional.conv1d(
+        torch.nn.functional.pad(x, (pqmf.pad, pqmf.pad)),
+        pqmf.analysis_filter, stride=subbands)
+    sub = fused.pqmf_analysis(pqmf, x)
+    _cmp(sub, sub_ref, what="pqmf analysis")
+
+    s0 = torch.randn(2, subbands, 1024, device=DEV)
+    n = subbands
+    up = torch.nn.functional.conv_transpose1d(
+        s0, pqmf.updown_filter * n, stride=n)
+    y_ref = torch.nn.functional.conv1d(
+        torch.nn.functional.pad(up, (pqmf.pad, pqmf.pad)),
+        pqmf.synthesis_filter.transpose(0, 1))
+    y = fused.pqmf_synthesis(pqmf, s0)
+    _cmp(y, y_ref, what="pqmf synthesis")
+
+    # gradient flow through synthesis
+    s1 = s0.clone().requires_grad_(True)
+    fused.pqmf_synthesis(pqmf, s1).pow(2).mean().backward()
+    s2 = s0.clone().requires_grad_(True)
+    up2 = torch.nn.functional.conv_transpose1d(
+        s2, pqmf.updown_filter * n, stride=n)
+    torch.nn.functional.conv1d(
+        torch.nn.functional.pad(up2, (pqmf.pad, pqmf.pad)),
+        pqmf.synthesis_filter.transpose(0, 1)).pow(2).mean().backward()
+    _cmp(s1.grad, s2.grad, rtol=1e-3, atol=1e-4, what="pqmf synth gx")
+
+
+def test_bf16_conv_close_to_fp32():
+    mod = _mkconv(256, 256, 3, pad_mode="reflect", act="leaky")
+    x = torch.randn(2, 256, 512, device=DEV)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    y32 = fused_conv1d(mod, x)
+    y16 = fused_conv1d(mod, x.to(torch.bfloat16))
+    assert y16.dtype == torch.bfloat16
+    rel = (y16.float() - y32).abs().mean() / y32.abs().mean()
+    assert rel < 2e-2, f"bf16 rel err {rel:.4f}"
+
+
+def test_fused_adam_gpu():
+    from melgan_amd.ops.adam import FusedAdam
+
+    torch.manual_seed(0)
+    p1 = torch.nn.Parameter(torch.randn(1000, 33, device=DEV))
+    p2 = torch.nn.Parameter(p1.detach().clone())
+    p3 = torch.nn.Parameter(torch.randn(77, device=DEV))
+    p4 = torch.nn.Parameter(p3.detach().clone())
+    o_ref = torch.optim.Adam([p1, p3], lr=1e-3, betas=(0.5, 0.9))
+    o_fused = FusedAdam([p2, p4], lr=1e-3, betas=(0.5, 0.9))
+    for i in range(5):
+        g1 = torch.randn_like(p1)
+        g3 = torch.randn_like(p3)
+        p1.grad = g1.clone()
+        p2.grad = g1.clone()
+        p3.grad = g3.clone()
+        p4.grad = g3.clone()
+        o_ref.step()
+        o_fused.step()
+    _cmp(p2, p1, rtol=1e-5, atol=1e-5, what="adam p")
+    _cmp(p4, p3, rtol=1e-5, atol=1e-5, what="adam p small")
+
+
+def test_generator_fused_vs_composite():
+    """Whole generator: HIP path vs composite path, same weights."""
+    import os
+
+    from melgan_amd.models.generator import Generator
+
+    torch.manual_seed(0)
+    g = Generator(80, 4, 128, [8, 4, 2], 3).to(DEV)
+    mel = torch.randn(2, 80, 32, device=DEV)
+    y_hip = g(mel)
+    os.environ["MELGAN_AMD_DISABLE_HIP"] = "1"
+    try:
+        y_ref = g(mel)
+    finally:
+        del os.environ["MELGAN_AMD_DISABLE_HIP"]
+    _cmp(y_hip, y_ref, rtol=1e-3, atol=1e-3, what="generator e2e")
+
+
+def test_train_step_gpu():
+    """One full multi-band GAN train step on the HIP path: finite losses."""
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.engine.trainer import Trainer
+
+    raw = """
+data: {train_dir: null, val_dir: null}
+train: {batch_size: 2, num_workers: 0, discriminator_train_start_steps: 0}
+audio: {segment_length: 8192, hop_length: 256}
+model: {out_channels: 4, base_channels: 128, generator_ratio: [8, 4, 2],
+        use_subband_stft_loss: true}
+log: {chkpt_dir: /tmp/gpu_ck, log_dir: /tmp/gpu_logs}
+backend: {amp: bf16, fused_adam: true}
+"""
+    hp = merge_default(HParam(raw=raw))
+    tr = Trainer(hp, "gpu_smoke", device=torch.device(DEV), enable_tb=False)
+    torch.manual_seed(0)
+    mel = torch.randn(2, 80, 32, device=DEV)
+    audio = torch.rand(2, 1, 8192, device=DEV) * 1.6 - 0.8
+    for _ in range(3):
+        g_loss, parts, d_loss = tr.train_step(mel, audio)
+        assert torch.isfinite(g_loss)
+        assert d_loss is None or torch.isfinite(d_loss)
+
+
+def test_native_extension_is_loaded():
+    """Contract: on a GPU box the HIP extension must be the running path."""
+    import melgan_amd._C as C
+
+    assert hasattr(C, "conv1d_fwd")
+    from melgan_amd.ops import _ext
+
+    x = torch.randn(4, device=DEV)
+    assert _ext.use_hip(x)
