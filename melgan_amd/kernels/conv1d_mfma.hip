@@ -222,10 +222,11 @@ conv1d_bwd_data_mfma_kernel(const unsigned short* __restrict__ gz,
 }
 
 // un-pad fold: gx[j] = gxpad[p+j] (+ mirrored edge terms for reflect)
+// src_len >= Tin + 2*pad (the zero-stuffed bwd-data path over-computes)
 template <typename T>
 __global__ void unpad_fold_kernel(const T* __restrict__ gxpad,
                                   T* __restrict__ gx, int BC, int Tin,
-                                  int pad, bool reflect) {
+                                  int pad, bool reflect, int src_len) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t n = (int64_t)BC * Tin;
   const int64_t step = (int64_t)gridDim.x * blockDim.x;
@@ -233,7 +234,7 @@ __global__ void unpad_fold_kernel(const T* __restrict__ gxpad,
   for (int64_t idx = i; idx < n; idx += step) {
     const int64_t bc = idx / Tin;
     const int j = (int)(idx - bc * Tin);
-    const T* src = gxpad + bc * Tpad;
+    const T* src = gxpad + bc * src_len;
     float v = static_cast<float>(src[pad + j]);
     if (reflect && pad > 0) {
       if (j > 0 && j <= pad) v += static_cast<float>(src[pad - j]);
@@ -246,10 +247,133 @@ __global__ void unpad_fold_kernel(const T* __restrict__ gxpad,
 }
 
 // ---------------------------------------------------------------------------
-// backward-weight (stride 1): GEMM  M=Cout, N=Cin*K, Kred=B*To
-//   gw[co][ci*K+k] += sum_{b,t} gz[b][co][t] * pre(x[b][ci][t + k*d - p])
-// grid: (ceil(N/64), ceil(Cout/64), NSPLIT); fp32 atomicAdd epilogue.
-// LDS: lds_gz[64co][72t] ++ lds_xs[64col][72t]
+// generic grouped forward (any stride / dilation / groups / K):
+//   y[b][co][t] = act(bias + sum_{cig,k} w[co][cig][k] *
+//                     pre(x[b][g*Cpg+cig][t*s + k*d - p]))
+// The contraction runs over FUSED (cig,k) columns in chunks of 32; the
+// shifted input rows are expanded into LDS per chunk.  Covers the
+// discriminator's grouped strided k=41 towers, k=15
+// (Cin=1), and — via zero-stuffing + weight flip on the host — strided
+// conv backward-data and conv_transpose1d forward (K2).
+// grid: (ceil(To/64), G * ceil(Copg/64), B)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+conv1d_fwd_mfma_g_kernel(const unsigned short* __restrict__ x,
+                         const unsigned short* __restrict__ w,
+                         const float* __restrict__ bias,
+                         unsigned short* __restrict__ y, int B, int Cin,
+                         int Tin, int Cout, int To, int K, int stride,
+                         int dil, int pad, bool reflect, int groups, int act,
+                         float slope, bool act_before) {
+  __shared__ __attribute__((aligned(16))) unsigned short smem[2 * 64 * LDW];
+  unsigned short* lds_w = smem;              // [co][kk]
+  unsigned short* lds_xs = smem + 64 * LDW;  // [col=t][kk]
+
+  const int Copg = Cout / groups;
+  const int Cpg = Cin / groups;
+  const int mtiles = (Copg + BM - 1) / BM;
+  const int g = blockIdx.y / mtiles;
+  const int mt = blockIdx.y - g * mtiles;
+  const int co_base = g * Copg + mt * BM;
+  const int co_end = (g + 1) * Copg;
+  const int t0 = blockIdx.x * BN;
+  const int b = blockIdx.z;
+  const int Npg = Cpg * K;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int row = lane & 15;
+  const int kk0 = (lane >> 4) * 8;
+
+  f32x4 acc[2][2] = {};
+
+  for (int nn0 = 0; nn0 < Npg; nn0 += CI_CH) {
+    for (int idx = tid; idx < BM * CI_CH; idx += 256) {
+      const int co = idx / CI_CH;
+      const int kk = idx - co * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (co_base + co < co_end && nn < Npg)
+        v = w[((int64_t)(co_base + co) * Cpg + nn / K) * K + nn % K];
+      lds_w[co * LDW + kk] = v;
+    }
+    for (int idx = tid; idx < BN * CI_CH; idx += 256) {
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (nn < Npg && t0 + col < To) {
+        const int cig = nn / K;
+        const int k = nn - cig * K;
+        const unsigned short* xr =
+            x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+        v = load_pad_bf16(xr, (t0 + col) * stride + k * dil - pad, Tin,
+                          reflect);
+        if (act_before) v = leaky_bf16(v, slope);
+      }
+      lds_xs[col * LDW + kk] = v;
+    }
+    __syncthreads();
+
+    {
+      bf16x8 a0 = *(const bf16x8*)(lds_w + (wr * 32 + row) * LDW + kk0);
+      bf16x8 a1 = *(const bf16x8*)(lds_w + (wr * 32 + 16 + row) * LDW + kk0);
+      bf16x8 b0 = *(const bf16x8*)(lds_xs + (wc * 32 + row) * LDW + kk0);
+      bf16x8 b1 =
+          *(const bf16x8*)(lds_xs + (wc * 32 + 16 + row) * LDW + kk0);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+  for (int m = 0; m < 2; ++m)
+    for (int n = 0; n < 2; ++n) {
+      const int t = t0 + wc * 32 + n * 16 + col;
+      if (t >= To) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co_base + wr * 32 + m * 16 + row4 + r;
+        if (co >= co_end || co >= Cout) continue;
+        float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
+        if (!act_before) o = apply_act(o, act, slope);
+        union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+        y[((int64_t)b * Cout + co) * To + t] = cv.s;
+      }
+    }
+}
+
+// zero-stuff along t: y[b][c][t*s] = x[b][c][t] (rest 0)
+__global__ void zero_stuff_kernel(const unsigned short* __restrict__ x,
+                                  unsigned short* __restrict__ y,
+                                  int64_t BC, int Tin, int stride,
+                                  int Lout) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t n = BC * Lout;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = i; idx < n; idx += step) {
+    const int64_t bc = idx / Lout;
+    const int t = (int)(idx - bc * Lout);
+    unsigned short v = 0;
+    if (t % stride == 0 && t / stride < Tin) v = x[bc * Tin + t / stride];
+    y[idx] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// generalized backward-weight: GEMM  M=Cout (grouped), N=Cpg*K, Kred=B*To
+//   gw[co][cig][k] += sum_{b,t} A[b][co][t] * pre(Bx[b][g*Cpg+cig][t*s+k*d-p])
+// act_a: LeakyReLU on the A operand (conv_transpose weight grad, where
+//        A is the pre-activation input x); act_b: LeakyReLU on B (conv
+//        with fused input activation).  Shared by conv1d (A=gz, B=x) and
+//        conv_transpose1d (A=x, B=gy, roles swapped by the wrapper).
+// grid: (ceil(Cpg*K/64), G*ceil(Copg/64), NSPLIT); fp32 atomicAdd epilogue.
 // ---------------------------------------------------------------------------
 constexpr int TCH = 64;   // t per chunk
 constexpr int LDT = 72;   // padded t stride (144 B rows)
@@ -258,16 +382,23 @@ __global__ void __launch_bounds__(256)
 conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
                               const unsigned short* __restrict__ x,
                               float* __restrict__ gw, int B, int Cin,
-                              int Tin, int Cout, int To, int K, int dil,
-                              int pad, bool reflect, float slope,
-                              bool act_before, int nsplit) {
+                              int Tin, int Cout, int To, int K, int stride,
+                              int dil, int pad, bool reflect, int groups,
+                              float slope, bool act_a, bool act_b,
+                              int nsplit) {
   __shared__ __attribute__((aligned(16))) unsigned short smem[2 * 64 * LDT];
   unsigned short* lds_gz = smem;            // [co][t]
-  unsigned short* lds_xs = smem + 64 * LDT;  // [col=(ci,k)][t]
+  unsigned short* lds_xs = smem + 64 * LDT;  // [col=(cig,k)][t]
 
-  const int n0 = blockIdx.x * 64;   // (ci,k) tile origin
-  const int co0 = blockIdx.y * 64;  // co tile origin
-  const int N = Cin * K;
+  const int Copg = Cout / groups;
+  const int Cpg = Cin / groups;
+  const int mtiles = (Copg + BM - 1) / BM;
+  const int g = blockIdx.y / mtiles;
+  const int mt = blockIdx.y - g * mtiles;
+  const int co_base = g * Copg + mt * BM;
+  const int co_end = (g + 1) * Copg;
+  const int n0 = blockIdx.x * 64;   // (cig,k) tile origin within the group
+  const int N = Cpg * K;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -286,8 +417,10 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
       const int co = idx / TCH;
       const int t = idx - co * TCH;
       unsigned short v = 0;
-      if (co0 + co < Cout && t0 + t < To)
-        v = gz[((int64_t)b * Cout + co0 + co) * To + t0 + t];
+      if (co_base + co < co_end && t0 + t < To) {
+        v = gz[((int64_t)b * Cout + co_base + co) * To + t0 + t];
+        if (act_a) v = leaky_bf16(v, slope);
+      }
       lds_gz[co * LDT + t] = v;
     }
     for (int idx = tid; idx < 64 * TCH; idx += 256) {
@@ -296,11 +429,12 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
       unsigned short v = 0;
       const int nn = n0 + cc;
       if (nn < N && t0 + t < To) {
-        const int ci = nn / K;
-        const int k = nn - ci * K;
+        const int ci = g * Cpg + nn / K;
+        const int k = nn % K;
         const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-        v = load_pad_bf16(xr, (t0 + t) + k * dil - pad, Tin, reflect);
-        if (act_before) v = leaky_bf16(v, slope);
+        v = load_pad_bf16(xr, (t0 + t) * stride + k * dil - pad, Tin,
+                          reflect);
+        if (act_b) v = leaky_bf16(v, slope);
       }
       lds_xs[cc * LDT + t] = v;
     }
@@ -327,13 +461,13 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
     for (int n = 0; n < 2; ++n) {
       const int nn = n0 + wc * 32 + n * 16 + col;
       if (nn >= N) continue;
-      const int ci = nn / K;
-      const int k = nn - ci * K;
+      const int cig = nn / K;
+      const int k = nn - cig * K;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int co = co0 + wr * 32 + m * 16 + row4 + r;
-        if (co >= Cout) continue;
-        atomicAdd(&gw[((int64_t)co * Cin + ci) * K + k], acc[m][n][r]);
+        const int co = co_base + wr * 32 + m * 16 + row4 + r;
+        if (co >= co_end || co >= Cout) continue;
+        atomicAdd(&gw[((int64_t)co * Cpg + cig) * K + k], acc[m][n][r]);
       }
     }
 }
@@ -349,6 +483,15 @@ static inline int cdiv_i(int64_t a, int64_t b) {
   return (int)((a + b - 1) / b);
 }
 
+bool mfma_enabled() {
+  static const bool disabled = []() {
+    const char* e = getenv("MELGAN_AMD_NO_MFMA");
+    return e && e[0] == '1';
+  }();
+  return !disabled;
+}
+
+// specialized stride-1 groups-1 kernel (x span staged once per ci-chunk)
 bool conv1d_mfma_eligible(const torch::Tensor& x, int64_t Cin, int64_t K,
                           int64_t stride, int64_t dil, int64_t groups) {
   if (x.scalar_type() != torch::kBFloat16) return false;
@@ -357,11 +500,12 @@ bool conv1d_mfma_eligible(const torch::Tensor& x, int64_t Cin, int64_t K,
   const int span = 64 + (int)((K - 1) * dil);
   const size_t lds = (size_t)(span + K * 64) * LDW * 2;
   if (lds > 64 * 1024) return false;
-  static const bool disabled = []() {
-    const char* e = getenv("MELGAN_AMD_NO_MFMA");
-    return e && e[0] == '1';
-  }();
-  return !disabled;
+  return mfma_enabled();
+}
+
+// generic grouped kernel: any stride/dil/groups/K for bf16
+bool conv1d_mfma_eligible_any(const torch::Tensor& x) {
+  return x.scalar_type() == torch::kBFloat16 && mfma_enabled();
 }
 
 torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
@@ -384,6 +528,42 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+torch::Tensor conv1d_fwd_mfma_generic(torch::Tensor x, torch::Tensor w,
+                                      const float* bias_p, int64_t stride,
+                                      int64_t dil, int64_t pad, bool reflect,
+                                      int64_t groups, int64_t act,
+                                      double slope, bool act_before,
+                                      int64_t To) {
+  const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
+  const int Cout = w.size(0), K = w.size(2);
+  const int Copg = Cout / (int)groups;
+  auto y = torch::empty({B, Cout, (int)To}, x.options());
+  dim3 grid(cdiv_i(To, BN), (int)groups * cdiv_i(Copg, BM), B);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(conv1d_fwd_mfma_g_kernel, grid, dim3(256), 0, stream,
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)w.data_ptr(), bias_p,
+                     (unsigned short*)y.data_ptr(), B, Cin, Tin, Cout,
+                     (int)To, K, (int)stride, (int)dil, (int)pad, reflect,
+                     (int)groups, (int)act, (float)slope, act_before);
+  return y;
+}
+
+static void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
+                           int Tin, int pad, bool reflect, int src_len) {
+  const int64_t n = (int64_t)BC * Tin;
+  const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, gx.scalar_type(), "unpad_fold", [&] {
+        hipLaunchKernelGGL((unpad_fold_kernel<scalar_t>), dim3(blocks),
+                           dim3(256), 0, stream,
+                           gxpad.data_ptr<scalar_t>(),
+                           gx.data_ptr<scalar_t>(), BC, Tin, pad, reflect,
+                           src_len);
+      });
+}
+
 torch::Tensor conv1d_bwd_data_mfma(torch::Tensor gz, torch::Tensor w,
                                    int64_t dil, int64_t pad, bool reflect,
                                    int64_t Tin) {
@@ -402,29 +582,68 @@ torch::Tensor conv1d_bwd_data_mfma(torch::Tensor gz, torch::Tensor w,
                      To, K, (int)dil);
   if (pad == 0) return gxpad;
   auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
-  const int64_t n = (int64_t)B * Cin * Tin;
-  const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);
-  AT_DISPATCH_FLOATING_TYPES_AND2(
-      at::kBFloat16, at::kHalf, gz.scalar_type(), "unpad_fold", [&] {
-        hipLaunchKernelGGL((unpad_fold_kernel<scalar_t>), dim3(blocks),
-                           dim3(256), 0, stream,
-                           gxpad.data_ptr<scalar_t>(),
-                           gx.data_ptr<scalar_t>(), B * Cin, (int)Tin,
-                           (int)pad, reflect);
-      });
+  run_unpad_fold(gxpad, gx, B * Cin, (int)Tin, (int)pad, reflect, Tpad);
+  return gx;
+}
+
+// generic bwd-data: zero-stuff gz (stride>1), flip/permute w, run the
+// grouped generic forward on padded coordinates, fold the padding.
+torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
+                                           torch::Tensor w, int64_t stride,
+                                           int64_t dil, int64_t pad,
+                                           bool reflect, int64_t groups,
+                                           int64_t Tin) {
+  const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
+  const int Cpg = w.size(1), K = w.size(2);
+  const int Cin = Cpg * (int)groups;
+  const int Copg = Cout / (int)groups;
+  auto stream = at::hip::getCurrentHIPStream();
+
+  // w~ [Cin][Copg][K] = w[g*Copg+cog][cig][K-1-k]
+  auto wf = w.view({(int)groups, Copg, Cpg, K})
+                .permute({0, 2, 1, 3})
+                .flip(-1)
+                .reshape({Cin, Copg, K})
+                .contiguous();
+
+  torch::Tensor gz_in = gz;
+  int Lin = To;
+  if (stride > 1) {
+    Lin = (To - 1) * (int)stride + 1;
+    gz_in = torch::empty({B, Cout, Lin}, gz.options());
+    const int64_t n = (int64_t)B * Cout * Lin;
+    const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);
+    hipLaunchKernelGGL(zero_stuff_kernel, dim3(blocks), dim3(256), 0,
+                       stream, (const unsigned short*)gz.data_ptr(),
+                       (unsigned short*)gz_in.data_ptr(),
+                       (int64_t)B * Cout, To, (int)stride, Lin);
+  }
+  // gxpad' = conv(gz_in, w~, stride 1, dil d, zero-pad (K-1)d)
+  const int pad2 = (K - 1) * (int)dil;
+  const int Lout = Lin + pad2;  // Lin + 2*pad2 - (K-1)*dil
+  const int Tpad = (int)Tin + 2 * (int)pad;
+  TORCH_CHECK(Lout >= Tpad, "bwd_data geometry: Lout < Tpad");
+  auto gxpad = conv1d_fwd_mfma_generic(gz_in, wf, nullptr, 1, dil, pad2,
+                                       false, groups, 0, 0.0, false, Lout);
+  auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
+  run_unpad_fold(gxpad, gx, B * Cin, (int)Tin, (int)pad, reflect, Lout);
   return gx;
 }
 
 torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
-                                     int64_t dil, int64_t pad, bool reflect,
-                                     int64_t K, bool act_before,
-                                     double slope) {
+                                     int64_t stride, int64_t dil,
+                                     int64_t pad, bool reflect,
+                                     int64_t groups, int64_t K, bool act_a,
+                                     bool act_b, double slope) {
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
-  const int N = Cin * (int)K;
-  auto gw = torch::zeros({Cout, Cin, (int)K},
+  const int Cpg = Cin / (int)groups;
+  const int Copg = Cout / (int)groups;
+  const int N = Cpg * (int)K;
+  auto gw = torch::zeros({Cout, Cpg, (int)K},
                          x.options().dtype(torch::kFloat));
-  const int gx_ = cdiv_i(N, 64), gy_ = cdiv_i(Cout, 64);
+  const int gx_ = cdiv_i(N, 64);
+  const int gy_ = (int)groups * cdiv_i(Copg, 64);
   const int tchunks = cdiv_i(To, TCH);
   const int64_t total = (int64_t)B * tchunks;
   int nsplit = (int)std::min<int64_t>(
@@ -435,7 +654,34 @@ torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
                      stream, (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
                      gw.data_ptr<float>(), B, Cin, Tin, Cout, To, (int)K,
-                     (int)dil, (int)pad, reflect, (float)slope, act_before,
-                     nsplit);
+                     (int)stride, (int)dil, (int)pad, reflect, (int)groups,
+                     (float)slope, act_a, act_b, nsplit);
   return gw;
+}
+
+// conv_transpose forward: zero-stuff x, flip/permute w, generic conv.
+torch::Tensor conv_transpose1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
+                                        const float* bias_p, int64_t stride,
+                                        int64_t pad, int64_t out_pad,
+                                        int64_t act, double slope,
+                                        bool act_before, int64_t To) {
+  const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
+  const int Cout = w.size(1), K = w.size(2);
+  auto stream = at::hip::getCurrentHIPStream();
+  // w' [Cout][Cin][k'] = w[ci][co][K-1-k']
+  auto wf = w.permute({1, 0, 2}).flip(-1).contiguous();
+  const int Lin = (Tin - 1) * (int)stride + 1 + (int)out_pad;
+  auto x_up = torch::empty({B, Cin, Lin}, x.options());
+  {
+    const int64_t n = (int64_t)B * Cin * Lin;
+    const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);
+    hipLaunchKernelGGL(zero_stuff_kernel, dim3(blocks), dim3(256), 0,
+                       stream, (const unsigned short*)x.data_ptr(),
+                       (unsigned short*)x_up.data_ptr(), (int64_t)B * Cin,
+                       Tin, (int)stride, Lin);
+  }
+  const int pad2 = K - 1 - (int)pad;
+  TORCH_CHECK(pad2 >= 0, "conv_transpose geometry: pad > K-1");
+  return conv1d_fwd_mfma_generic(x_up, wf, bias_p, 1, 1, pad2, false, 1,
+                                 act, slope, act_before, To);
 }

@@ -361,20 +361,38 @@ using namespace melgan;
 
 static inline int cdiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
-// MFMA tier (conv1d_mfma.hip) — stride-1 groups-1 bf16 implicit GEMM
+// MFMA tier (conv1d_mfma.hip) — bf16 implicit GEMM
 bool conv1d_mfma_eligible(const torch::Tensor& x, int64_t Cin, int64_t K,
                           int64_t stride, int64_t dil, int64_t groups);
+bool conv1d_mfma_eligible_any(const torch::Tensor& x);
 torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                               const float* bias_p, int64_t dil, int64_t pad,
                               bool reflect, int64_t act, double slope,
                               bool act_before, int64_t To);
+torch::Tensor conv1d_fwd_mfma_generic(torch::Tensor x, torch::Tensor w,
+                                      const float* bias_p, int64_t stride,
+                                      int64_t dil, int64_t pad, bool reflect,
+                                      int64_t groups, int64_t act,
+                                      double slope, bool act_before,
+                                      int64_t To);
 torch::Tensor conv1d_bwd_data_mfma(torch::Tensor gz, torch::Tensor w,
                                    int64_t dil, int64_t pad, bool reflect,
                                    int64_t Tin);
+torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
+                                           torch::Tensor w, int64_t stride,
+                                           int64_t dil, int64_t pad,
+                                           bool reflect, int64_t groups,
+                                           int64_t Tin);
 torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
-                                     int64_t dil, int64_t pad, bool reflect,
-                                     int64_t K, bool act_before,
-                                     double slope);
+                                     int64_t stride, int64_t dil,
+                                     int64_t pad, bool reflect,
+                                     int64_t groups, int64_t K, bool act_a,
+                                     bool act_b, double slope);
+torch::Tensor conv_transpose1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
+                                        const float* bias_p, int64_t stride,
+                                        int64_t pad, int64_t out_pad,
+                                        int64_t act, double slope,
+                                        bool act_before, int64_t To);
 
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
@@ -398,6 +416,9 @@ torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
   if (conv1d_mfma_eligible(x, Cin, K, stride, dil, groups))
     return conv1d_fwd_mfma(x, w, bias_p, dil, pad, reflect, act, slope,
                            act_before, To);
+  if (conv1d_mfma_eligible_any(x))
+    return conv1d_fwd_mfma_generic(x, w, bias_p, stride, dil, pad, reflect,
+                                   groups, act, slope, act_before, To);
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(To, BT), cdiv(Cout, BC), B);
   auto stream = at::hip::getCurrentHIPStream();
@@ -422,6 +443,9 @@ torch::Tensor conv1d_bwd_data(torch::Tensor gz, torch::Tensor w,
   const int Cin = w.size(1) * groups, K = w.size(2);
   if (conv1d_mfma_eligible(gz, Cout, K, stride, dil, groups))
     return conv1d_bwd_data_mfma(gz, w, dil, pad, reflect, Tin);
+  if (conv1d_mfma_eligible_any(gz))
+    return conv1d_bwd_data_mfma_generic(gz, w, stride, dil, pad, reflect,
+                                        groups, Tin);
   auto gx = torch::empty({B, (int)Cin, (int)Tin}, gz.options());
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(Tin, BT), cdiv(Cin, BC), B);
@@ -447,9 +471,10 @@ torch::Tensor conv1d_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
   const int Cpg = Cin / (int)groups;
-  if (conv1d_mfma_eligible(x, Cin, K, stride, dil, groups))
-    return conv1d_bwd_weight_mfma(gz, x, dil, pad, reflect, K, act_before,
-                                  slope);
+  if (conv1d_mfma_eligible_any(x))
+    return conv1d_bwd_weight_mfma(gz, x, stride, dil, pad, reflect, groups,
+                                  K, /*act_a=*/false,
+                                  /*act_b=*/act_before, slope);
   auto gw = torch::empty({Cout, Cpg, (int)K},
                          x.options().dtype(torch::kFloat));
   // chunk (b,t) so small weights still fill the chip
@@ -483,13 +508,16 @@ torch::Tensor conv_transpose1d_fwd(torch::Tensor x, torch::Tensor w,
   const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
   const int Cout = w.size(1), K = w.size(2);
   const int To = (Tin - 1) * (int)stride - 2 * (int)pad + K + (int)out_pad;
-  auto y = torch::empty({B, Cout, To}, x.options());
   torch::Tensor bias_f;
   const float* bias_p = nullptr;
   if (bias.has_value()) {
     bias_f = bias->to(torch::kFloat).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
+  if (conv1d_mfma_eligible_any(x))
+    return conv_transpose1d_fwd_mfma(
+        x, w, bias_p, stride, pad, out_pad, ACT_NONE, slope, act_before, To);
+  auto y = torch::empty({B, Cout, To}, x.options());
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(To, BT), cdiv(Cout, BC), B);
   auto stream = at::hip::getCurrentHIPStream();
@@ -514,6 +542,11 @@ torch::Tensor conv_transpose1d_bwd_data(torch::Tensor gy, torch::Tensor w,
   CHECK_IN(w);
   const int B = gy.size(0), Cout = gy.size(1), To = gy.size(2);
   const int Cin = w.size(0), K = w.size(2);
+  // gx[ci][m] = direct conv of gy with weight viewed (Cin, Cout, K):
+  // reuse the conv1d_fwd dispatcher (MFMA generic path for bf16).
+  if (conv1d_mfma_eligible_any(gy))
+    return conv1d_fwd_mfma_generic(gy, w, nullptr, stride, 1, pad, false, 1,
+                                   ACT_NONE, 0.0, false, Tin);
   auto gx = torch::empty({B, Cin, (int)Tin}, gy.options());
   constexpr int BT = 64, BC = 4;
   dim3 grid(cdiv(Tin, BT), cdiv(Cin, BC), B);
@@ -538,6 +571,13 @@ torch::Tensor conv_transpose1d_bwd_weight(torch::Tensor gy, torch::Tensor x,
   CHECK_IN(x);
   const int B = gy.size(0), Cout = gy.size(1), To = gy.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
+  // gw[ci][co][k] = sum_{b,m} pre(x[ci][m]) * gy[co][m*s - p + k]:
+  // the generalized MFMA weight-grad kernel with roles swapped
+  // (A = x [act_a], B = gy), dil=1.
+  if (conv1d_mfma_eligible_any(x))
+    return conv1d_bwd_weight_mfma(x, gy, stride, 1, pad, false, 1, K,
+                                  /*act_a=*/act_before, /*act_b=*/false,
+                                  slope);
   auto gw = torch::empty({Cin, Cout, (int)K},
                          x.options().dtype(torch::kFloat));
   int64_t bm = (int64_t)B * Tin;
