@@ -606,10 +606,13 @@ torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
                 .reshape({Cin, Copg, K})
                 .contiguous();
 
+  const int pad2 = (K - 1) * (int)dil;
+  const int Tpad = (int)Tin + 2 * (int)pad;
   torch::Tensor gz_in = gz;
   int Lin = To;
   if (stride > 1) {
-    Lin = (To - 1) * (int)stride + 1;
+    // cover Tpad: trailing zeros beyond the last real sample are harmless
+    Lin = std::max((To - 1) * (int)stride + 1, Tpad - pad2);
     gz_in = torch::empty({B, Cout, Lin}, gz.options());
     const int64_t n = (int64_t)B * Cout * Lin;
     const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);
@@ -619,9 +622,7 @@ torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
                        (int64_t)B * Cout, To, (int)stride, Lin);
   }
   // gxpad' = conv(gz_in, w~, stride 1, dil d, zero-pad (K-1)d)
-  const int pad2 = (K - 1) * (int)dil;
   const int Lout = Lin + pad2;  // Lin + 2*pad2 - (K-1)*dil
-  const int Tpad = (int)Tin + 2 * (int)pad;
   TORCH_CHECK(Lout >= Tpad, "bwd_data geometry: Lout < Tpad");
   auto gxpad = conv1d_fwd_mfma_generic(gz_in, wf, nullptr, 1, dil, pad2,
                                        false, groups, 0, 0.0, false, Lout);
