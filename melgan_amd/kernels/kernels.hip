@@ -291,18 +291,22 @@ __global__ void leaky_bwd_inplace_kernel(T* __restrict__ gx,
   }
 }
 
-// gbias[c] = sum_{b,t} gz[b][c][t];  grid: (C), block 256, LDS tree
+// gbias[c] = sum_{b,t} gz[b][c][t];  grid: (C, chunks), atomic combine
 template <typename T>
 __global__ void bias_grad_kernel(const T* __restrict__ gz,
                                  float* __restrict__ gb, int B, int C,
-                                 int Tlen) {
+                                 int Tlen, int nchunks) {
   __shared__ float red[256];
   const int c = blockIdx.x;
+  const int64_t bt_total = (int64_t)B * Tlen;
+  const int64_t per = (bt_total + nchunks - 1) / nchunks;
+  const int64_t lo = (int64_t)blockIdx.y * per;
+  const int64_t hi = min(lo + per, bt_total);
   float acc = 0.f;
-  for (int b = 0; b < B; ++b) {
-    const T* row = gz + ((int64_t)b * C + c) * Tlen;
-    for (int t = threadIdx.x; t < Tlen; t += blockDim.x)
-      acc += static_cast<float>(row[t]);
+  for (int64_t bt = lo + threadIdx.x; bt < hi; bt += blockDim.x) {
+    const int b = (int)(bt / Tlen);
+    const int t = (int)(bt - (int64_t)b * Tlen);
+    acc += static_cast<float>(gz[((int64_t)b * C + c) * Tlen + t]);
   }
   red[threadIdx.x] = acc;
   __syncthreads();
@@ -310,7 +314,74 @@ __global__ void bias_grad_kernel(const T* __restrict__ gz,
     if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
     __syncthreads();
   }
-  if (threadIdx.x == 0) gb[c] = red[0];
+  if (threadIdx.x == 0) {
+    if (nchunks == 1) gb[c] = red[0];
+    else atomicAdd(&gb[c], red[0]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fused weight-norm fold / backward (K5 elimination):
+//   wn_fold:  n[c] = max(||v[c]||, 1e-12); w[c] = v[c] * g[c] / n[c]
+//   wn_bwd:   dot = <gw[c], v[c]>; gg[c] = dot/n;
+//             gv[c] = (g/n) gw[c] - (g dot / n^3) v[c]
+// One kernel launch each, replacing the ~6-kernel torch chains per conv.
+// grid: (C0); block 256.
+// ---------------------------------------------------------------------------
+template <typename TO>
+__global__ void wn_fold_kernel(const float* __restrict__ v,
+                               const float* __restrict__ g,
+                               TO* __restrict__ w, float* __restrict__ n_out,
+                               int C0, int M) {
+  __shared__ float red[256];
+  const int c = blockIdx.x;
+  const float* vr = v + (int64_t)c * M;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < M; i += blockDim.x) {
+    const float vv = vr[i];
+    acc += vv * vv;
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  const float n = fmaxf(sqrtf(red[0]), 1e-12f);
+  const float scale = g[c] / n;
+  if (threadIdx.x == 0) n_out[c] = n;
+  TO* wr = w + (int64_t)c * M;
+  for (int i = threadIdx.x; i < M; i += blockDim.x)
+    wr[i] = static_cast<TO>(vr[i] * scale);
+}
+
+__global__ void wn_bwd_kernel(const float* __restrict__ gw,
+                              const float* __restrict__ v,
+                              const float* __restrict__ g,
+                              const float* __restrict__ n,
+                              float* __restrict__ gv,
+                              float* __restrict__ gg, int C0, int M) {
+  __shared__ float red[256];
+  const int c = blockIdx.x;
+  const float* gwr = gw + (int64_t)c * M;
+  const float* vr = v + (int64_t)c * M;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < M; i += blockDim.x)
+    acc += gwr[i] * vr[i];
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  const float dot = red[0];
+  const float nc = n[c];
+  const float a = g[c] / nc;
+  const float bcoef = g[c] * dot / (nc * nc * nc);
+  if (threadIdx.x == 0) gg[c] = dot / nc;
+  float* gvr = gv + (int64_t)c * M;
+  for (int i = threadIdx.x; i < M; i += blockDim.x)
+    gvr[i] = a * gwr[i] - bcoef * vr[i];
 }
 
 // ---------------------------------------------------------------------------
@@ -634,15 +705,57 @@ void leaky_bwd_inplace(torch::Tensor gx, torch::Tensor x, double slope) {
 torch::Tensor conv1d_bias_grad(torch::Tensor gz) {
   CHECK_IN(gz);
   const int B = gz.size(0), C = gz.size(1), Tlen = gz.size(2);
-  auto gb = torch::empty({C}, gz.options().dtype(torch::kFloat));
+  const int nchunks = (int)std::min<int64_t>(
+      std::max<int64_t>(1, (int64_t)B * Tlen / 16384),
+      std::max<int64_t>(1, 2048 / C));
+  auto gb = nchunks == 1
+                ? torch::empty({C}, gz.options().dtype(torch::kFloat))
+                : torch::zeros({C}, gz.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, gz.scalar_type(), "bias_grad", [&] {
-        hipLaunchKernelGGL((bias_grad_kernel<scalar_t>), dim3(C), dim3(256),
-                           0, stream, gz.data_ptr<scalar_t>(),
-                           gb.data_ptr<float>(), B, C, Tlen);
+        hipLaunchKernelGGL((bias_grad_kernel<scalar_t>), dim3(C, nchunks),
+                           dim3(256), 0, stream, gz.data_ptr<scalar_t>(),
+                           gb.data_ptr<float>(), B, C, Tlen, nchunks);
       });
   return gb;
+}
+
+std::vector<torch::Tensor> wn_fold(torch::Tensor v, torch::Tensor g,
+                                   torch::ScalarType out_dtype) {
+  CHECK_IN(v);
+  TORCH_CHECK(v.scalar_type() == torch::kFloat, "wn_fold: fp32 v");
+  const int C0 = v.size(0);
+  const int M = (int)(v.numel() / C0);
+  auto gc = g.to(torch::kFloat).flatten().contiguous();
+  auto w = torch::empty(v.sizes(), v.options().dtype(out_dtype));
+  auto n = torch::empty({C0}, v.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, out_dtype, "wn_fold", [&] {
+        hipLaunchKernelGGL((wn_fold_kernel<scalar_t>), dim3(C0), dim3(256),
+                           0, stream, v.data_ptr<float>(),
+                           gc.data_ptr<float>(), w.data_ptr<scalar_t>(),
+                           n.data_ptr<float>(), C0, M);
+      });
+  return {w, n};
+}
+
+std::vector<torch::Tensor> wn_bwd(torch::Tensor gw, torch::Tensor v,
+                                  torch::Tensor g, torch::Tensor n) {
+  CHECK_IN(gw);
+  CHECK_IN(v);
+  const int C0 = v.size(0);
+  const int M = (int)(v.numel() / C0);
+  auto gc = g.to(torch::kFloat).flatten().contiguous();
+  auto gv = torch::empty_like(v);
+  auto gg = torch::empty({C0}, v.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(wn_bwd_kernel, dim3(C0), dim3(256), 0, stream,
+                     gw.data_ptr<float>(), v.data_ptr<float>(),
+                     gc.data_ptr<float>(), n.data_ptr<float>(),
+                     gv.data_ptr<float>(), gg.data_ptr<float>(), C0, M);
+  return {gv, gg};
 }
 
 void multi_tensor_adam(std::vector<torch::Tensor> params,
@@ -697,5 +810,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_bwd", &act_bwd);
   m.def("leaky_bwd_inplace", &leaky_bwd_inplace);
   m.def("conv1d_bias_grad", &conv1d_bias_grad);
+  m.def("wn_fold", &wn_fold);
+  m.def("wn_bwd", &wn_bwd);
   m.def("multi_tensor_adam", &multi_tensor_adam);
 }

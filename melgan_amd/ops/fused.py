@@ -30,34 +30,37 @@ from melgan_amd.ops import _ext
 from melgan_amd.ops.wnconv import ACT_LEAKY, ACT_NONE, ACT_TANH
 
 
-def _wn_weight_and_norm(v: torch.Tensor, g: torch.Tensor):
-    n = v.flatten(1).norm(dim=1).clamp_min(1e-12)  # (C0,)
-    shape = (-1,) + (1,) * (v.dim() - 1)
-    w = v * (g.flatten() / n).reshape(shape)
+def _wn_fold(v: torch.Tensor, g: torch.Tensor, dtype: torch.dtype):
+    """(w cast to dtype, n fp32) — ONE fused kernel (K5 elimination)."""
+    C = _ext.require_ext()
+    w, n = C.wn_fold(v.contiguous(), g, dtype)
+    return w, n
+
+
+def _wn_cached(mod, dtype: torch.dtype):
+    """Fold w=g*v/||v|| once per optimizer step (param _version keyed)."""
+    key = (mod.weight_v._version, mod.weight_g._version, dtype)
+    cache = getattr(mod, "_wn_cache", None)
+    if cache is not None and cache[0] == key:
+        return cache[1], cache[2]
+    w, n = _wn_fold(mod.weight_v, mod.weight_g, dtype)
+    mod._wn_cache = (key, w, n)
     return w, n
 
 
 def _wn_backward(gw: torch.Tensor, v: torch.Tensor, g: torch.Tensor,
                  n: torch.Tensor):
-    """grads of w = g*v/n wrt v and g; all fp32."""
-    gw = gw.float()
-    v = v.float()
-    gflat = g.flatten().float()
-    shape = (-1,) + (1,) * (v.dim() - 1)
-    dot = (gw * v).flatten(1).sum(1)  # (C0,)
-    grad_g = (dot / n).reshape(g.shape)
-    grad_v = (gflat / n).reshape(shape) * gw \
-        - ((gflat * dot) / (n ** 3)).reshape(shape) * v
-    return grad_v, grad_g
+    """grads of w = g*v/n wrt v and g; fused kernel, fp32."""
+    C = _ext.require_ext()
+    gv, gg = C.wn_bwd(gw.contiguous(), v.contiguous(), g, n)
+    return gv, gg.reshape(g.shape)
 
 
 class FusedWNConv1dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, v, g, bias, stride, dilation, pad, reflect, groups,
-                act, slope, act_before):
+    def forward(ctx, x, v, g, bias, w_c, n, stride, dilation, pad, reflect,
+                groups, act, slope, act_before):
         C = _ext.require_ext()
-        w, n = _wn_weight_and_norm(v.float(), g.float())
-        w_c = w.to(x.dtype)
         b_c = bias.to(x.dtype) if bias is not None else None
         y = C.conv1d_fwd(x, w_c, b_c, stride, dilation, pad, reflect,
                          groups, act, slope, act_before)
@@ -90,7 +93,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
         if ctx.has_bias:
             gb = C.conv1d_bias_grad(gz)
         return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
-                None, None, None, None, None)
+                None, None, None, None, None, None, None)
 
 
 def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
@@ -105,18 +108,18 @@ def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
         return C.conv1d_fwd(x, w_c, b_c, mod.stride, mod.dilation, mod.pad,
                             mod.pad_mode == "reflect", mod.groups, mod.act,
                             mod.act_slope, mod.act_before)
+    w_c, n = _wn_cached(mod, x.dtype)
     return FusedWNConv1dFn.apply(
-        x, mod.weight_v, mod.weight_g, mod.bias, mod.stride, mod.dilation,
-        mod.pad, mod.pad_mode == "reflect", mod.groups, mod.act,
-        mod.act_slope, mod.act_before)
+        x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, mod.stride,
+        mod.dilation, mod.pad, mod.pad_mode == "reflect", mod.groups,
+        mod.act, mod.act_slope, mod.act_before)
 
 
 class FusedWNConvT1dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, v, g, bias, stride, pad, out_pad, slope, act_before):
+    def forward(ctx, x, v, g, bias, w_c, n, stride, pad, out_pad, slope,
+                act_before):
         C = _ext.require_ext()
-        w, n = _wn_weight_and_norm(v.float(), g.float())  # (Cin, Cout, K)
-        w_c = w.to(x.dtype)
         b_c = bias.to(x.dtype) if bias is not None else None
         y = C.conv_transpose1d_fwd(x, w_c, b_c, stride, pad, out_pad,
                                    slope, act_before)
@@ -144,7 +147,7 @@ class FusedWNConvT1dFn(torch.autograd.Function):
         if ctx.has_bias:
             gb = C.conv1d_bias_grad(gy)
         return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
-                None, None)
+                None, None, None, None)
 
 
 def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
@@ -158,9 +161,10 @@ def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
         return C.conv_transpose1d_fwd(x, w_c, b_c, mod.stride, mod.padding,
                                       mod.output_padding, mod.act_slope,
                                       mod.act_before)
+    w_c, n = _wn_cached(mod, x.dtype)
     return FusedWNConvT1dFn.apply(
-        x, mod.weight_v, mod.weight_g, mod.bias, mod.stride, mod.padding,
-        mod.output_padding, mod.act_slope, mod.act_before)
+        x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, mod.stride,
+        mod.padding, mod.output_padding, mod.act_slope, mod.act_before)
 
 
 # ---------------------------------------------------------------------------
