@@ -440,6 +440,104 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// phase-decomposed strided backward-data (zero-pad geometry only):
+// outputs j with (j+p) % s == phi form one GEMM per phase:
+//   gx[b][ci][j=u*s+phi-p] = sum_{co in grp, i} w[co][cig][phi+i*s] *
+//                            gz[b][co][u-i]
+// No zero-stuffing (s x less MFMA work than the stuffed path) and no
+// host-side weight flip.  16-row M tile (grouped convs have Cpg<=16).
+// grid: (s * ceil(Umax/128), G * ceil(Cpg/16), B)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
+                        const unsigned short* __restrict__ w,
+                        unsigned short* __restrict__ gx, int B, int Cin,
+                        int Tin, int Cout, int To, int K, int stride,
+                        int pad, int groups, int utiles) {
+  __shared__ __attribute__((aligned(16))) unsigned short smem[
+      16 * LDW + 128 * LDW];
+  unsigned short* lds_w = smem;              // [16 cig][kk]
+  unsigned short* lds_gs = smem + 16 * LDW;  // [col=u(128)][kk]
+
+  const int Copg = Cout / groups;
+  const int Cpg = Cin / groups;
+  const int mtiles = (Cpg + 15) / 16;
+  const int g = blockIdx.y / mtiles;
+  const int mt = blockIdx.y - g * mtiles;
+  const int ci_base = g * Cpg + mt * 16;
+  const int ci_end = (g + 1) * Cpg;
+  const int phi = blockIdx.x / utiles;       // output phase
+  const int u0 = (blockIdx.x - phi * utiles) * 128;
+  const int b = blockIdx.z;
+  const int nk = (K - phi + stride - 1) / stride;  // taps for this phase
+  const int Npg = Copg * nk;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int row = lane & 15;
+  const int kk0 = (lane >> 4) * 8;
+
+  f32x4 acc[2] = {};
+
+  for (int nn0 = 0; nn0 < Npg; nn0 += CI_CH) {
+    for (int idx = tid; idx < 16 * CI_CH; idx += 256) {
+      const int cig = idx / CI_CH;
+      const int kk = idx - cig * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (ci_base + cig < ci_end && nn < Npg) {
+        const int co = nn / nk;
+        const int i = nn - co * nk;
+        const int k = phi + i * stride;
+        v = w[((int64_t)(g * Copg + co) * Cpg + (ci_base + cig) -
+               g * Cpg) * K + k];
+      }
+      lds_w[cig * LDW + kk] = v;
+    }
+    for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (nn < Npg) {
+        const int co = nn / nk;
+        const int i = nn - co * nk;
+        const int t = u0 + col - i;
+        if (t >= 0 && t < To)
+          v = gz[((int64_t)b * Cout + g * Copg + co) * To + t];
+      }
+      lds_gs[col * LDW + kk] = v;
+    }
+    __syncthreads();
+    {
+      bf16x8 a0 = *(const bf16x8*)(lds_w + row * LDW + kk0);
+      bf16x8 b0 = *(const bf16x8*)(lds_gs + (wid * 32 + row) * LDW + kk0);
+      bf16x8 b1 =
+          *(const bf16x8*)(lds_gs + (wid * 32 + 16 + row) * LDW + kk0);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+  for (int n = 0; n < 2; ++n) {
+    const int u = u0 + wid * 32 + n * 16 + col;
+    const int j = u * stride + phi - pad;
+    if (j < 0 || j >= Tin) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int ci = ci_base + row4 + r;
+      if (ci >= ci_end || ci >= Cin) continue;
+      union { __bf16 bb; unsigned short s; } cv{(__bf16)acc[n][r]};
+      gx[((int64_t)b * Cin + ci) * Tin + j] = cv.s;
+    }
+  }
+}
+
 // zero-stuff along t: y[b][c][t*s] = x[b][c][t] (rest 0)
 __global__ void zero_stuff_kernel(const unsigned short* __restrict__ x,
                                   unsigned short* __restrict__ y,
@@ -709,6 +807,23 @@ torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
 
   const int pad2 = (K - 1) * (int)dil;
   const int Tpad = (int)Tin + 2 * (int)pad;
+  const int CpgX = Cpg;  // (per-group input channels == gx channels/group)
+  if (!reflect && stride > 1 && dil == 1 && CpgX <= 16) {
+    auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
+    // Umax covers all j in [0, Tin): u <= (Tin - 1 + pad)/stride
+    const int Umax = ((int)Tin - 1 + (int)pad) / (int)stride + 1;
+    const int utiles = cdiv_i(Umax, 128);
+    dim3 grid((int)stride * utiles,
+              (int)groups * cdiv_i(CpgX, 16), B);
+    auto stream2 = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(convbwd_phased16_kernel, grid, dim3(256), 0, stream2,
+                       (const unsigned short*)gz.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (unsigned short*)gx.data_ptr(), B, Cin, (int)Tin,
+                       Cout, To, K, (int)stride, (int)pad, (int)groups,
+                       utiles);
+    return gx;
+  }
   torch::Tensor gz_in = gz;
   int Lin = To;
   if (stride > 1) {

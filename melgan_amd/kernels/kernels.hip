@@ -800,6 +800,22 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
                      (float)beta2, (float)eps, (float)wd, bc1, bc2);
 }
 
+void multi_tensor_adam_tbl(torch::Tensor tbl, torch::Tensor chunks,
+                           int64_t nt, double lr, double beta1, double beta2,
+                           double eps, double wd, int64_t step) {
+  TORCH_CHECK(tbl.is_cuda() && chunks.is_cuda());
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  auto stream = at::hip::getCurrentHIPStream();
+  int64_t* tp = tbl.data_ptr<int64_t>();
+  hipLaunchKernelGGL(multi_tensor_adam_kernel,
+                     dim3((int)chunks.numel()), dim3(256), 0, stream,
+                     (float**)tp, (float**)(tp + nt), (float**)(tp + 2 * nt),
+                     (float**)(tp + 3 * nt), tp + 4 * nt,
+                     chunks.data_ptr<int64_t>(), (float)lr, (float)beta1,
+                     (float)beta2, (float)eps, (float)wd, bc1, bc2);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_fwd", &conv1d_fwd);
   m.def("conv1d_bwd_data", &conv1d_bwd_data);
@@ -813,4 +829,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wn_fold", &wn_fold);
   m.def("wn_bwd", &wn_bwd);
   m.def("multi_tensor_adam", &multi_tensor_adam);
+  m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

@@ -51,11 +51,12 @@ class FusedAdam(torch.optim.Optimizer):
             beta1, beta2 = group["betas"]
             e = _ext.ext()
             if e is not None and params[0].is_cuda and hasattr(
-                    e, "multi_tensor_adam"):
-                e.multi_tensor_adam(
-                    params, grads, exp_avgs, exp_avg_sqs,
-                    group["lr"], beta1, beta2, group["eps"],
-                    group["weight_decay"], steps[0])
+                    e, "multi_tensor_adam_tbl"):
+                tbl, chunks, nt = self._tables(
+                    id(group), params, grads, exp_avgs, exp_avg_sqs)
+                e.multi_tensor_adam_tbl(
+                    tbl, chunks, nt, group["lr"], beta1, beta2,
+                    group["eps"], group["weight_decay"], steps[0])
             else:
                 # torch's own multi-tensor path (same math)
                 torch._foreach_mul_(exp_avgs, beta1)
@@ -71,3 +72,31 @@ class FusedAdam(torch.optim.Optimizer):
                         p.mul_(1 - group["lr"] * group["weight_decay"])
                     p.addcdiv_(m, denom, value=-group["lr"] / bc1)
         return loss
+
+    def _tables(self, key, params, grads, exp_avgs, exp_avg_sqs):
+        """Device pointer tables, cached across steps (rebuilt only when a
+        pointer changes, e.g. grads reallocated). Keeps the step at ONE
+        kernel launch with no per-step H2D, and is hipGraph-capture safe
+        once pointers are static."""
+        import torch as _t
+
+        CHUNK = 1 << 16
+        ptrs = []
+        for lst in (params, grads, exp_avgs, exp_avg_sqs):
+            ptrs.extend(t.data_ptr() for t in lst)
+        cache = getattr(self, "_tbl_cache", {}).get(key)
+        if cache is not None and cache[0] == ptrs:
+            return cache[1], cache[2], cache[3]
+        nt = len(params)
+        host = ptrs + [p.numel() for p in params]
+        chunks = []
+        for i, p in enumerate(params):
+            for off in range(0, p.numel(), CHUNK):
+                chunks.append((i << 48) | off)
+        dev = params[0].device
+        tbl = _t.tensor(host, dtype=_t.long).to(dev)
+        ch = _t.tensor(chunks, dtype=_t.long).to(dev)
+        if not hasattr(self, "_tbl_cache"):
+            self._tbl_cache = {}
+        self._tbl_cache[key] = (ptrs, tbl, ch, nt)
+        return tbl, ch, nt
