@@ -22,6 +22,7 @@ OUT_SO = os.path.join(PKG_DIR, "_C.so")
 SOURCES = [
     "kernels.hip",
     "conv1d_mfma.hip",
+    "stft.hip",
 ]
 
 

@@ -465,6 +465,15 @@ torch::Tensor conv_transpose1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                                         int64_t act, double slope,
                                         bool act_before, int64_t To);
 
+// STFT pipeline (stft.hip)
+torch::Tensor stft_frame(torch::Tensor x, int64_t fft, int64_t hop,
+                         int64_t win, torch::Tensor window);
+torch::Tensor stft_frame_bwd(torch::Tensor gframes, torch::Tensor window,
+                             int64_t fft, int64_t hop, int64_t win,
+                             int64_t S);
+torch::Tensor complex_mag_clamp(torch::Tensor spec_ri, double mn);
+torch::Tensor mag_bwd(torch::Tensor gmag, torch::Tensor spec_ri, double mn);
+
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
                          int64_t dil, int64_t pad, bool reflect,
@@ -828,6 +837,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_bias_grad", &conv1d_bias_grad);
   m.def("wn_fold", &wn_fold);
   m.def("wn_bwd", &wn_bwd);
+  m.def("stft_frame", &stft_frame);
+  m.def("stft_frame_bwd", &stft_frame_bwd);
+  m.def("complex_mag_clamp", &complex_mag_clamp);
+  m.def("mag_bwd", &mag_bwd);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

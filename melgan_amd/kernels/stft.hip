@@ -1,0 +1,202 @@
+// Fused STFT-loss pipeline pieces on gfx950.
+//
+// torch.stft on ROCm lowers to ~6 kernels per call (reflect-pad
+// materialisation, as_strided copy, window mul, rocFFT, abs, clamp).
+// Here the framing prologue (reflect pad + window, no materialised pad)
+// and the |.|+clamp epilogue are ONE kernel each; the batched R2C FFT
+// itself stays on rocFFT via torch.fft.rfft between them (its autograd
+// is handled by torch).  All fp32 (GAN loss numerics).
+//
+//   stft_frame(x, fft, hop, win, window) -> (B, F, fft) frames
+//   stft_frame_bwd(gframes, window, ...) -> gx  (overlap-add gather +
+//                                           reflect un-pad fold)
+//   complex_mag_clamp(spec_ri, min)      -> |S| clamped
+//   mag_bwd(gmag, spec_ri, mag, min)     -> gspec_ri
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace melgan {
+
+// frames[b][f][j] = wpad[j] * xpad[b][f*hop + j]
+// xpad = reflect-pad(x, fft/2);  wpad = window centered in fft
+__global__ void stft_frame_kernel(const float* __restrict__ x,
+                                  const float* __restrict__ window,
+                                  float* __restrict__ frames, int B, int S,
+                                  int F, int fft, int hop, int win) {
+  const int off = (fft - win) / 2;
+  const int pad = fft / 2;
+  const int64_t n = (int64_t)B * F * fft;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const int j = (int)(i % fft);
+    const int64_t bf = i / fft;
+    const int f = (int)(bf % F);
+    const int b = (int)(bf / F);
+    float v = 0.f;
+    const int jw = j - off;
+    if (jw >= 0 && jw < win) {
+      const int q = f * hop + j - pad;  // un-padded coordinate
+      v = window[jw] * x[(int64_t)b * S + reflect_idx(q, S)];
+    }
+    frames[i] = v;
+  }
+}
+
+// gxpad[b][q] = sum_{f: 0<=q-f*hop<fft} gframes[b][f][q-f*hop] * wpad[...]
+__global__ void stft_frame_bwd_kernel(const float* __restrict__ gframes,
+                                      const float* __restrict__ window,
+                                      float* __restrict__ gxpad, int B,
+                                      int Spad, int F, int fft, int hop,
+                                      int win) {
+  const int off = (fft - win) / 2;
+  const int64_t n = (int64_t)B * Spad;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const int q = (int)(i % Spad);
+    const int b = (int)(i / Spad);
+    // contributing frames: f*hop <= q < f*hop + fft
+    const int f_lo = (q >= fft) ? ((q - fft) / hop + 1) : 0;
+    float acc = 0.f;
+    for (int f = f_lo; f <= q / hop && f < F; ++f) {
+      const int j = q - f * hop;
+      const int jw = j - off;
+      if (jw >= 0 && jw < win)
+        acc += gframes[((int64_t)b * F + f) * fft + j] * window[jw];
+    }
+    gxpad[i] = acc;
+  }
+}
+
+__global__ void mag_clamp_kernel(const float* __restrict__ spec_ri,
+                                 float* __restrict__ mag, int64_t n,
+                                 float mn) {
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const float re = spec_ri[2 * i];
+    const float im = spec_ri[2 * i + 1];
+    mag[i] = fmaxf(sqrtf(re * re + im * im), mn);
+  }
+}
+
+// d|S|/dS = S/|S|; zero where the clamp was active (|S| < mn)
+__global__ void mag_bwd_kernel(const float* __restrict__ gmag,
+                               const float* __restrict__ spec_ri,
+                               float* __restrict__ gspec_ri, int64_t n,
+                               float mn) {
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const float re = spec_ri[2 * i];
+    const float im = spec_ri[2 * i + 1];
+    const float r = sqrtf(re * re + im * im);
+    float s = 0.f;
+    if (r >= mn && r > 0.f) s = gmag[i] / r;
+    gspec_ri[2 * i] = re * s;
+    gspec_ri[2 * i + 1] = im * s;
+  }
+}
+
+// declared in conv1d_mfma.hip
+template <typename T>
+__global__ void unpad_fold_stft_kernel(const T* __restrict__ gxpad,
+                                       T* __restrict__ gx, int BC, int Tin,
+                                       int pad, bool reflect, int src_len) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t n = (int64_t)BC * Tin;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  const int Tpad = Tin + 2 * pad;
+  for (int64_t idx = i; idx < n; idx += step) {
+    const int64_t bc = idx / Tin;
+    const int j = (int)(idx - bc * Tin);
+    const T* src = gxpad + bc * src_len;
+    float v = static_cast<float>(src[pad + j]);
+    if (reflect && pad > 0) {
+      if (j > 0 && j <= pad) v += static_cast<float>(src[pad - j]);
+      const int jm = 2 * Tin - 2 - j;
+      if (jm >= Tin && (pad + jm) < Tpad)
+        v += static_cast<float>(src[pad + jm]);
+    }
+    gx[idx] = static_cast<T>(v);
+  }
+}
+
+}  // namespace melgan
+
+using namespace melgan;
+
+static inline int gblocks(int64_t n) {
+  return (int)std::min<int64_t>(2048, (n + 255) / 256);
+}
+
+torch::Tensor stft_frame(torch::Tensor x, int64_t fft, int64_t hop,
+                         int64_t win, torch::Tensor window) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kFloat);
+  const int B = x.size(0), S = x.size(1);
+  const int F = S / (int)hop + 1;  // center=True frame count
+  auto frames = torch::empty({B, F, (int)fft}, x.options());
+  auto wc = window.contiguous();
+  auto stream = at::hip::getCurrentHIPStream();
+  const int64_t n = (int64_t)B * F * fft;
+  hipLaunchKernelGGL(stft_frame_kernel, dim3(gblocks(n)), dim3(256), 0,
+                     stream, x.data_ptr<float>(), wc.data_ptr<float>(),
+                     frames.data_ptr<float>(), B, S, F, (int)fft, (int)hop,
+                     (int)win);
+  return frames;
+}
+
+torch::Tensor stft_frame_bwd(torch::Tensor gframes, torch::Tensor window,
+                             int64_t fft, int64_t hop, int64_t win,
+                             int64_t S) {
+  TORCH_CHECK(gframes.is_cuda() && gframes.is_contiguous());
+  const int B = gframes.size(0), F = gframes.size(1);
+  const int pad = (int)fft / 2;
+  const int Spad = (int)S + 2 * pad;
+  auto gxpad = torch::empty({B, Spad},
+                            gframes.options().dtype(torch::kFloat));
+  auto wc = window.contiguous();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(stft_frame_bwd_kernel,
+                     dim3(gblocks((int64_t)B * Spad)), dim3(256), 0, stream,
+                     gframes.data_ptr<float>(), wc.data_ptr<float>(),
+                     gxpad.data_ptr<float>(), B, Spad, F, (int)fft,
+                     (int)hop, (int)win);
+  auto gx = torch::empty({B, (int)S}, gxpad.options());
+  hipLaunchKernelGGL(unpad_fold_stft_kernel<float>,
+                     dim3(gblocks((int64_t)B * S)), dim3(256), 0, stream,
+                     gxpad.data_ptr<float>(), gx.data_ptr<float>(), B,
+                     (int)S, pad, true, Spad);
+  return gx;
+}
+
+torch::Tensor complex_mag_clamp(torch::Tensor spec_ri, double mn) {
+  TORCH_CHECK(spec_ri.is_cuda() && spec_ri.is_contiguous());
+  auto sizes = spec_ri.sizes().vec();
+  TORCH_CHECK(sizes.back() == 2);
+  sizes.pop_back();
+  auto mag = torch::empty(sizes, spec_ri.options());
+  const int64_t n = mag.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mag_clamp_kernel, dim3(gblocks(n)), dim3(256), 0,
+                     stream, spec_ri.data_ptr<float>(),
+                     mag.data_ptr<float>(), n, (float)mn);
+  return mag;
+}
+
+torch::Tensor mag_bwd(torch::Tensor gmag, torch::Tensor spec_ri, double mn) {
+  TORCH_CHECK(gmag.is_cuda() && gmag.is_contiguous());
+  auto gspec = torch::empty_like(spec_ri);
+  const int64_t n = gmag.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mag_bwd_kernel, dim3(gblocks(n)), dim3(256), 0, stream,
+                     gmag.data_ptr<float>(), spec_ri.data_ptr<float>(),
+                     gspec.data_ptr<float>(), n, (float)mn);
+  return gspec;
+}

@@ -231,42 +231,52 @@ def pqmf_synthesis(mod, x: torch.Tensor) -> torch.Tensor:
 # Fused STFT magnitude (frame+window prologue, |.| epilogue on rocFFT)
 # ---------------------------------------------------------------------------
 def fused_stft_mag(x, fft_size, hop_size, win_length, window):
+    """|STFT| with fused framing prologue and magnitude epilogue around
+    rocFFT (torch.fft.rfft keeps its exact autograd vjp).  Falls back to
+    torch.stft when the extension lacks the kernels."""
     C = _ext.ext()
     if C is None or not hasattr(C, "stft_frame"):
-        # rocFFT via torch.stft — still fully on-GPU
         xs = torch.stft(x.float(), fft_size, hop_size, win_length,
                         window=window.float(), return_complex=True,
                         center=True, pad_mode="reflect")
         return torch.clamp(xs.abs().transpose(1, 2), min=1e-7)
-    return _FusedSTFTMagFn.apply(x, fft_size, hop_size, win_length, window)
+    xf = x.float().contiguous()
+    frames = _FrameFn.apply(xf, fft_size, hop_size, win_length,
+                            window.float())
+    spec = torch.fft.rfft(frames, dim=-1)  # rocFFT batched R2C
+    return _MagClampFn.apply(spec)
 
 
-class _FusedSTFTMagFn(torch.autograd.Function):
+class _FrameFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, fft_size, hop_size, win_length, window):
         C = _ext.require_ext()
-        xf = x.float().contiguous()
-        frames = C.stft_frame(xf, fft_size, hop_size, win_length,
-                              window.float())  # (B, n_frames, fft)
-        spec = torch.fft.rfft(frames, dim=-1)  # rocFFT batched R2C
-        mag = C.complex_mag_clamp(torch.view_as_real(spec), 1e-7)
-        ctx.save_for_backward(torch.view_as_real(spec), mag, window)
         ctx.geom = (fft_size, hop_size, win_length, x.size(-1))
+        ctx.save_for_backward(window)
+        return C.stft_frame(x, fft_size, hop_size, win_length, window)
+
+    @staticmethod
+    def backward(ctx, gframes):
+        C = _ext.require_ext()
+        (window,) = ctx.saved_tensors
+        fft_size, hop_size, win_length, S = ctx.geom
+        gx = C.stft_frame_bwd(gframes.contiguous(), window, fft_size,
+                              hop_size, win_length, S)
+        return gx, None, None, None, None
+
+
+class _MagClampFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, spec):
+        C = _ext.require_ext()
+        spec_ri = torch.view_as_real(spec).contiguous()
+        mag = C.complex_mag_clamp(spec_ri, 1e-7)
+        ctx.save_for_backward(spec_ri)
         return mag
 
     @staticmethod
     def backward(ctx, gmag):
         C = _ext.require_ext()
-        spec_ri, mag, window = ctx.saved_tensors
-        fft_size, hop_size, win_length, t_in = ctx.geom
-        # d|S| backward: gS = gmag * S / |S| (fp32)
-        gspec_ri = C.mag_bwd(gmag.contiguous(), spec_ri, mag)
-        gspec = torch.view_as_complex(gspec_ri)
-        # rfft vjp: double the doubled bins then irfft-style adjoint
-        n = fft_size
-        gspec = gspec.clone()
-        gspec[..., 1:(n + 1) // 2] *= 2.0
-        gframes = torch.fft.irfft(gspec.conj(), n=n, dim=-1) * n
-        gx = C.stft_frame_bwd(gframes.contiguous(), window.float(),
-                              hop_size, win_length, t_in)
-        return gx, None, None, None, None
+        (spec_ri,) = ctx.saved_tensors
+        gspec_ri = C.mag_bwd(gmag.contiguous(), spec_ri, 1e-7)
+        return torch.view_as_complex(gspec_ri)
