@@ -28,6 +28,7 @@ from melgan_amd.ops.pqmf import PQMF
 from melgan_amd.ops.stft_loss import MultiResolutionSTFTLoss
 from melgan_amd.parallel.dp import BucketAllReduceDP
 from melgan_amd.config import check_hp_compat
+from melgan_amd.utils.profiling import PhaseTimer
 
 
 def make_optimizers(hp, generator, discriminator, device):
@@ -89,6 +90,9 @@ class Trainer:
                                     enable_tb=enable_tb, rank=rank)
         self.lambda_adv = hp.train.get("lambda_adv") or 2.5
         self.lambda_feat = hp.model.feat_match
+        self.timer = PhaseTimer(enabled=bool(hp.backend.get("phase_timing"))
+                                and self.device.type == "cuda")
+        self.last_phases = {}
 
     # ------------------------------------------------------------------
     def load(self, chkpt_path: str):
@@ -161,25 +165,37 @@ class Trainer:
         mel = mel.to(self.device, non_blocking=True)
         audio = audio.to(self.device, non_blocking=True)
 
+        tm = self.timer
         # ---- generator ------------------------------------------------
+        tm.mark("g_fwd")
         self.optim_g.zero_grad(set_to_none=True)
         g_loss, parts, fake = self._forward_g(mel, audio)
+        tm.mark("g_bwd")
         g_loss.backward()
+        tm.mark("g_comm_wait")
         self.dp_g.finish()
+        tm.mark("g_opt")
         self.optim_g.step()
 
         # ---- discriminator --------------------------------------------
         d_loss = None
         if self.step >= hp.train.discriminator_train_start_steps:
+            tm.mark("d_fwd")
             self.optim_d.zero_grad(set_to_none=True)
             with torch.autocast("cuda", dtype=self.amp_dtype,
                                 enabled=self.use_amp):
                 d_real = self.discriminator(audio.float())
                 d_fake = self.discriminator(fake.detach())
             d_loss = L.discriminator_loss(d_real, d_fake)
+            tm.mark("d_bwd")
             d_loss.backward()
+            tm.mark("d_comm_wait")
             self.dp_d.finish()
+            tm.mark("d_opt")
             self.optim_d.step()
+        tm.mark(None)
+        if tm.enabled:
+            self.last_phases = tm.collect()
 
         self.step += 1
         return g_loss.detach(), parts, \
@@ -210,7 +226,8 @@ class Trainer:
                 self.writer.log_training(
                     self.step, g_loss=g_loss, d_loss=d_loss,
                     stft=parts["stft"], sub=parts["sub"], adv=parts["adv"],
-                    feat=parts["feat"], mel_frames_per_sec=mfps)
+                    feat=parts["feat"], mel_frames_per_sec=mfps,
+                    **{f"ms_{k}": v for k, v in self.last_phases.items()})
                 if self.logger:
                     self.logger.info(
                         f"step {self.step} g_loss {float(g_loss):.4f}"

@@ -114,6 +114,32 @@ class Generator(nn.Module):
             layer.remove_weight_norm()
 
     @torch.no_grad()
+    def inference_chunked(self, mel: torch.Tensor, chunk_frames: int = 512,
+                          overlap: int = 16) -> torch.Tensor:
+        """Streaming inference for multi-minute audio:
+        process the mel in chunks with `overlap` frames of context on
+        each side and stitch by exact overlap-discard (the generator is
+        fully convolutional; `overlap` >= its receptive field in frames,
+        which is < 10 for every shipped config).  Memory is O(chunk)
+        instead of O(utterance)."""
+        if mel.dim() == 2:
+            mel = mel.unsqueeze(0)
+        T = mel.size(2)
+        if T <= chunk_frames + 2 * overlap:
+            return self.forward(mel)
+        hop_sub = self.hop_length // self.out_channels
+        outs = []
+        for i0 in range(0, T, chunk_frames):
+            i1 = min(i0 + chunk_frames, T)
+            lo = max(0, i0 - overlap)
+            hi = min(T, i1 + overlap)
+            y = self.forward(mel[:, :, lo:hi])
+            left = (i0 - lo) * hop_sub
+            right = (hi - i1) * hop_sub
+            outs.append(y[:, :, left:y.size(2) - right])
+        return torch.cat(outs, dim=2)
+
+    @torch.no_grad()
     def inference(self, mel: torch.Tensor, pad_frames: int = 10) -> torch.Tensor:
         """Family-style inference helper: pad the mel with
         `pad_frames` frames of silence on each side, run forward, trim the

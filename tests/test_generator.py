@@ -80,3 +80,17 @@ def test_param_count_order():
     g = Generator(80, 4, 512, [8, 4, 2], 3)
     n = sum(p.numel() for p in g.parameters())
     assert 1e6 < n < 6e6, n
+
+
+def test_chunked_inference_matches_full(tiny_mb_hp):
+    """Overlap-discard streaming == full forward."""
+    g = build_generator(tiny_mb_hp)
+    g.eval()
+    torch.manual_seed(0)
+    mel = torch.randn(1, 80, 300)
+    with torch.no_grad():
+        full = g(mel)
+        chunked = g.inference_chunked(mel, chunk_frames=64, overlap=16)
+    assert chunked.shape == full.shape
+    assert torch.allclose(chunked, full, atol=1e-5), \
+        (chunked - full).abs().max().item()
