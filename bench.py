@@ -128,7 +128,8 @@ def main():
             "dtype": "bf16" if have_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "mb-melgan-4band-22k",
+                "model": f"mb-melgan-{hp.model.out_channels}band-"
+                         f"{hp.audio.sampling_rate // 1000}k",
                 "global_batch": B * world,
                 "seq_len": frames,
                 "parallelism": f"dp{world}",
@@ -137,7 +138,8 @@ def main():
             },
         }
         if rtf is not None:
-            out["config"]["inference_rtf_22k"] = rtf
+            out["config"][
+                f"inference_rtf_{hp.audio.sampling_rate // 1000}k"] = rtf
         print(json.dumps(out))
 
 
