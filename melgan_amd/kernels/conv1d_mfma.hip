@@ -395,21 +395,28 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
         v = w[((int64_t)(co_base + co) * Cpg + nn / K) * K + nn % K];
       lds_w[co * LDW + kk] = v;
     }
-    for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
-      const int col = idx / CI_CH;
-      const int kk = idx - col * CI_CH;
-      const int nn = nn0 + kk;
-      unsigned short v = 0;
-      if (nn < Npg && t0 + col < To) {
-        const int cig = nn / K;
-        const int k = nn - cig * K;
-        const unsigned short* xr =
-            x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
-        v = load_pad_bf16(xr, (t0 + col) * stride + k * dil - pad, Tin,
-                          reflect);
-        if (act_before) v = leaky_bf16(v, slope);
+    for (int idx = tid; idx < 128 * (CI_CH / 4); idx += 256) {
+      const int col = idx / (CI_CH / 4);
+      const int kk4 = (idx - col * (CI_CH / 4)) * 4;
+      ushort4 v = {0, 0, 0, 0};
+      if (t0 + col < To) {
+        const int tpos = (t0 + col) * stride - pad;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int nn = nn0 + kk4 + u;
+          unsigned short e = 0;
+          if (nn < Npg) {
+            const int cig = nn / K;
+            const int k = nn - cig * K;
+            const unsigned short* xr =
+                x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+            e = load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
+            if (act_before) e = leaky_bf16(e, slope);
+          }
+          (&v.x)[u] = e;
+        }
       }
-      lds_xs[col * LDW + kk] = v;
+      *(ushort4*)(lds_xs + col * LDW + kk4) = v;
     }
     __syncthreads();
     {
@@ -602,30 +609,66 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
   for (int64_t c = blockIdx.z; c < total; c += nsplit) {
     const int b = (int)(c / tchunks);
     const int t0 = (int)(c - (int64_t)b * tchunks) * TCH;
-    for (int idx = tid; idx < 64 * TCH; idx += 256) {
-      const int co = idx / TCH;
-      const int t = idx - co * TCH;
-      unsigned short v = 0;
-      if (co_base + co < co_end && t0 + t < To) {
-        v = gz[((int64_t)b * Cout + co_base + co) * To + t0 + t];
-        if (act_a) v = leaky_bf16(v, slope);
+    // A stage (gz rows, always t-contiguous): ushort4 when the whole
+    // 64-t run is interior, scalar tail otherwise
+    const bool a_vec = (t0 + TCH <= To) && !act_a;
+    if (a_vec) {
+      for (int idx = tid; idx < 64 * (TCH / 4); idx += 256) {
+        const int co = idx / (TCH / 4);
+        const int t4 = (idx - co * (TCH / 4)) * 4;
+        ushort4 v = {0, 0, 0, 0};
+        if (co_base + co < co_end)
+          v = *(const ushort4*)(gz +
+                ((int64_t)b * Cout + co_base + co) * To + t0 + t4);
+        *(ushort4*)(lds_gz + co * LDT + t4) = v;
       }
-      lds_gz[co * LDT + t] = v;
+    } else {
+      for (int idx = tid; idx < 64 * TCH; idx += 256) {
+        const int co = idx / TCH;
+        const int t = idx - co * TCH;
+        unsigned short v = 0;
+        if (co_base + co < co_end && t0 + t < To) {
+          v = gz[((int64_t)b * Cout + co_base + co) * To + t0 + t];
+          if (act_a) v = leaky_bf16(v, slope);
+        }
+        lds_gz[co * LDT + t] = v;
+      }
     }
-    for (int idx = tid; idx < 64 * TCH; idx += 256) {
-      const int cc = idx / TCH;
-      const int t = idx - cc * TCH;
-      unsigned short v = 0;
+    // B stage (shifted x rows): vectorize per-column when the run is
+    // interior, 8-byte aligned and stride-1
+    for (int idx = tid; idx < 64 * (TCH / 4); idx += 256) {
+      const int cc = idx / (TCH / 4);
+      const int t4 = (idx - cc * (TCH / 4)) * 4;
       const int nn = n0 + cc;
-      if (nn < N && t0 + t < To) {
+      ushort4 v = {0, 0, 0, 0};
+      if (nn < N) {
         const int ci = g * Cpg + nn / K;
         const int k = nn % K;
         const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-        v = load_pad_bf16(xr, (t0 + t) * stride + k * dil - pad, Tin,
-                          reflect);
-        if (act_b) v = leaky_bf16(v, slope);
+        const int base = (t0 + t4) * stride + k * dil - pad;
+        if (stride == 1 && !reflect && (base & 3) == 0 && base >= 0 &&
+            base + 4 <= Tin && t0 + t4 + 4 <= To) {
+          v = *(const ushort4*)(xr + base);
+          if (act_b) {
+            v.x = leaky_bf16(v.x, slope);
+            v.y = leaky_bf16(v.y, slope);
+            v.z = leaky_bf16(v.z, slope);
+            v.w = leaky_bf16(v.w, slope);
+          }
+        } else {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            unsigned short e = 0;
+            if (t0 + t4 + u < To) {
+              e = load_pad_bf16(xr, (t0 + t4 + u) * stride + k * dil - pad,
+                                Tin, reflect);
+              if (act_b) e = leaky_bf16(e, slope);
+            }
+            (&v.x)[u] = e;
+          }
+        }
       }
-      lds_xs[cc * LDT + t] = v;
+      *(ushort4*)(lds_xs + cc * LDT + t4) = v;
     }
     __syncthreads();
 
