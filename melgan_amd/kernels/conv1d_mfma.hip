@@ -395,28 +395,21 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
         v = w[((int64_t)(co_base + co) * Cpg + nn / K) * K + nn % K];
       lds_w[co * LDW + kk] = v;
     }
-    for (int idx = tid; idx < 128 * (CI_CH / 4); idx += 256) {
-      const int col = idx / (CI_CH / 4);
-      const int kk4 = (idx - col * (CI_CH / 4)) * 4;
-      ushort4 v = {0, 0, 0, 0};
-      if (t0 + col < To) {
-        const int tpos = (t0 + col) * stride - pad;
-#pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          const int nn = nn0 + kk4 + u;
-          unsigned short e = 0;
-          if (nn < Npg) {
-            const int cig = nn / K;
-            const int k = nn - cig * K;
-            const unsigned short* xr =
-                x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
-            e = load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
-            if (act_before) e = leaky_bf16(e, slope);
-          }
-          (&v.x)[u] = e;
-        }
+    for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (nn < Npg && t0 + col < To) {
+        const int cig = nn / K;
+        const int k = nn - cig * K;
+        const unsigned short* xr =
+            x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+        v = load_pad_bf16(xr, (t0 + col) * stride + k * dil - pad, Tin,
+                          reflect);
+        if (act_before) v = leaky_bf16(v, slope);
       }
-      *(ushort4*)(lds_xs + col * LDW + kk4) = v;
+      lds_xs[col * LDW + kk] = v;
     }
     __syncthreads();
     {
@@ -542,6 +535,143 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
       union { __bf16 bb; unsigned short s; } cv{(__bf16)acc[n][r]};
       gx[((int64_t)b * Cin + ci) * Tin + j] = cv.s;
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// deep-contraction / small-T forward (disc k=5 etc.):
+// batch folded into the column dimension (N = B*To), 256-column tiles,
+// fused (ci,k) contraction — 4x more output per staging round than the
+// 64-column kernels, which are staging/barrier-bound at these shapes.
+// stride 1, groups 1.  grid: (ceil(B*To/256), ceil(Cout/64), 1)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
+                            const unsigned short* __restrict__ w,
+                            const float* __restrict__ bias,
+                            unsigned short* __restrict__ y, int B, int Cin,
+                            int Tin, int Cout, int To, int K, int dil,
+                            int pad, bool reflect, int act, float slope,
+                            bool act_before) {
+  __shared__ __attribute__((aligned(16))) unsigned short smem[
+      (64 + 256) * LDW];
+  unsigned short* lds_w = smem;               // [64co][kk]
+  unsigned short* lds_xs = smem + 64 * LDW;   // [col][kk]
+
+  const int cg0 = blockIdx.x * 256;           // global (b,t) column origin
+  const int co0 = blockIdx.y * BM;
+  const int NK = Cin * K;
+  const int64_t NCOL = (int64_t)B * To;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1, wc = wid & 1;      // wc: 128-col half
+  const int row = lane & 15;
+  const int kk0 = (lane >> 4) * 8;
+
+  f32x4 acc[2][8] = {};
+
+  for (int nn0 = 0; nn0 < NK; nn0 += CI_CH) {
+    for (int idx = tid; idx < BM * CI_CH; idx += 256) {
+      const int co = idx / CI_CH;
+      const int kk = idx - co * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (co0 + co < Cout && nn < NK)
+        v = w[((int64_t)(co0 + co) * Cin + nn / K) * K + nn % K];
+      lds_w[co * LDW + kk] = v;
+    }
+    for (int idx = tid; idx < 256 * CI_CH; idx += 256) {
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      const int64_t cg = cg0 + col;
+      if (nn < NK && cg < NCOL) {
+        const int b = (int)(cg / To);
+        const int t = (int)(cg - (int64_t)b * To);
+        const int ci = nn / K;
+        const int k = nn - ci * K;
+        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+        v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
+        if (act_before) v = leaky_bf16(v, slope);
+      }
+      lds_xs[col * LDW + kk] = v;
+    }
+    __syncthreads();
+
+    {
+      bf16x8 a0 = *(const bf16x8*)(lds_w + (wr * 32 + row) * LDW + kk0);
+      bf16x8 a1 = *(const bf16x8*)(lds_w + (wr * 32 + 16 + row) * LDW + kk0);
+#pragma unroll
+      for (int n = 0; n < 8; ++n) {
+        bf16x8 bn = *(const bf16x8*)(lds_xs +
+                                     (wc * 128 + n * 16 + row) * LDW + kk0);
+        acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bn, acc[0][n], 0, 0, 0);
+        acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bn, acc[1][n], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+  for (int m = 0; m < 2; ++m)
+    for (int n = 0; n < 8; ++n) {
+      const int64_t cg = cg0 + wc * 128 + n * 16 + col;
+      if (cg >= NCOL) continue;
+      const int b = (int)(cg / To);
+      const int t = (int)(cg - (int64_t)b * To);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co0 + wr * 32 + m * 16 + row4 + r;
+        if (co >= Cout) continue;
+        float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
+        if (!act_before) o = apply_act(o, act, slope);
+        union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+        y[((int64_t)b * Cout + co) * To + t] = cv.s;
+      }
+    }
+}
+
+// dot-form forward for tiny Cout (disc score conv 1024->1 k3): one block
+// per (b, t-tile); 256 threads reduce over Cin*K via LDS tree.
+__global__ void __launch_bounds__(256)
+conv1d_fwd_dot_kernel(const unsigned short* __restrict__ x,
+                      const unsigned short* __restrict__ w,
+                      const float* __restrict__ bias,
+                      unsigned short* __restrict__ y, int B, int Cin,
+                      int Tin, int Cout, int To, int K, int dil, int pad,
+                      bool reflect, int act, float slope, bool act_before) {
+  __shared__ float red[256];
+  const int t = blockIdx.x;
+  const int co = blockIdx.y;
+  const int b = blockIdx.z;
+  const int NK = Cin * K;
+  float acc = 0.f;
+  for (int nn = threadIdx.x; nn < NK; nn += 256) {
+    const int ci = nn / K;
+    const int k = nn - ci * K;
+    const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+    unsigned short v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
+    if (act_before) v = leaky_bf16(v, slope);
+    union { unsigned short s; __bf16 bb; } xv{v};
+    union { unsigned short s; __bf16 bb; } wv{
+        w[((int64_t)co * Cin + ci) * K + k]};
+    acc = fmaf((float)wv.bb, (float)xv.bb, acc);
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    float o = red[0] + (bias ? bias[co] : 0.f);
+    if (!act_before) o = apply_act(o, act, slope);
+    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+    y[((int64_t)b * Cout + co) * To + t] = cv.s;
   }
 }
 
@@ -747,6 +877,29 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
   const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
   const int Cout = w.size(0), K = w.size(2);
   auto y = torch::empty({B, Cout, (int)To}, x.options());
+  auto stream0 = at::hip::getCurrentHIPStream();
+  if (Cout <= 4 && Cin >= 256) {
+    // tiny-M: dot-form (one block per output sample)
+    dim3 grid((int)To, Cout, B);
+    hipLaunchKernelGGL(conv1d_fwd_dot_kernel, grid, dim3(256), 0, stream0,
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(), bias_p,
+                       (unsigned short*)y.data_ptr(), B, Cin, Tin, Cout,
+                       (int)To, K, (int)dil, (int)pad, reflect, (int)act,
+                       (float)slope, act_before);
+    return y;
+  }
+  if ((int64_t)Cin * K >= 2048 && To <= 256 && Cout >= 32) {
+    // deep contraction, small T: fold batch into 256-column tiles
+    dim3 grid(cdiv_i((int64_t)B * To, 256), cdiv_i(Cout, BM), 1);
+    hipLaunchKernelGGL(conv1d_fwd_mfma_deep_kernel, grid, dim3(256), 0,
+                       stream0, (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(), bias_p,
+                       (unsigned short*)y.data_ptr(), B, Cin, Tin, Cout,
+                       (int)To, K, (int)dil, (int)pad, reflect, (int)act,
+                       (float)slope, act_before);
+    return y;
+  }
   const int span = 64 + (K - 1) * (int)dil;
   const size_t lds = (size_t)(span + K * 64) * LDW * sizeof(unsigned short);
   dim3 grid(cdiv_i(To, BN), cdiv_i(Cout, BM), B);
@@ -791,7 +944,7 @@ torch::Tensor conv1d_fwd_mfma_generic(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
-static void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
+void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
                            int Tin, int pad, bool reflect, int src_len) {
   const int64_t n = (int64_t)BC * Tin;
   const int blocks = (int)std::min<int64_t>(2048, (n + 255) / 256);

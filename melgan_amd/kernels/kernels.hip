@@ -436,6 +436,8 @@ static inline int cdiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 bool conv1d_mfma_eligible(const torch::Tensor& x, int64_t Cin, int64_t K,
                           int64_t stride, int64_t dil, int64_t groups);
 bool conv1d_mfma_eligible_any(const torch::Tensor& x);
+void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC, int Tin,
+                    int pad, bool reflect, int src_len);
 torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                               const float* bias_p, int64_t dil, int64_t pad,
                               bool reflect, int64_t act, double slope,
@@ -521,6 +523,20 @@ torch::Tensor conv1d_bwd_data(torch::Tensor gz, torch::Tensor w,
   CHECK_IN(w);
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = w.size(1) * groups, K = w.size(2);
+  if (conv1d_mfma_eligible_any(gz) && stride == 1 && groups == 1 &&
+      (int64_t)Cout * K >= 2048 && Tin + 2 * pad <= 256 && Cin >= 32) {
+    // deep-contraction small-T (disc k5, mel conv): flipped-weight direct
+    // conv on the batch-folded 256-column kernel, then un-pad fold
+    auto wf = w.permute({1, 0, 2}).flip(-1).contiguous();
+    const int Tpad = (int)Tin + 2 * (int)pad;
+    auto gxpad = conv1d_fwd_mfma(gz, wf, nullptr, dil, (K - 1) * (int)dil,
+                                 false, 0, 0.0, false, Tpad);
+    if (pad == 0) return gxpad;
+    auto gx = torch::empty({B, (int)Cin, (int)Tin}, gz.options());
+    run_unpad_fold(gxpad, gx, B * (int)Cin, (int)Tin, (int)pad, reflect,
+                   Tpad);
+    return gx;
+  }
   if (conv1d_mfma_eligible(gz, Cout, K, stride, dil, groups))
     return conv1d_bwd_data_mfma(gz, w, dil, pad, reflect, Tin);
   if (conv1d_mfma_eligible_any(gz))
