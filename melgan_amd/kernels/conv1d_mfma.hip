@@ -549,7 +549,8 @@ __global__ void __launch_bounds__(256)
 conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
                             const unsigned short* __restrict__ w,
                             const float* __restrict__ bias,
-                            unsigned short* __restrict__ y, int B, int Cin,
+                            unsigned short* __restrict__ y,
+                            float* __restrict__ yws, int B, int Cin,
                             int Tin, int Cout, int To, int K, int dil,
                             int pad, bool reflect, int act, float slope,
                             bool act_before) {
@@ -572,7 +573,15 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
 
   f32x4 acc[2][8] = {};
 
-  for (int nn0 = 0; nn0 < NK; nn0 += CI_CH) {
+  // split-K over gridDim.z: this block covers chunk indices
+  // [z*per, min((z+1)*per, nchunks))
+  const int nchunks = (NK + CI_CH - 1) / CI_CH;
+  const int per = (nchunks + gridDim.z - 1) / gridDim.z;
+  const int c_lo = blockIdx.z * per;
+  const int c_hi = min(c_lo + per, nchunks);
+
+  for (int cix = c_lo; cix < c_hi; ++cix) {
+    const int nn0 = cix * CI_CH;
     for (int idx = tid; idx < BM * CI_CH; idx += 256) {
       const int co = idx / CI_CH;
       const int kk = idx - co * CI_CH;
@@ -627,12 +636,35 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
       for (int r = 0; r < 4; ++r) {
         const int co = co0 + wr * 32 + m * 16 + row4 + r;
         if (co >= Cout) continue;
-        float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
-        if (!act_before) o = apply_act(o, act, slope);
-        union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-        y[((int64_t)b * Cout + co) * To + t] = cv.s;
+        const int64_t oi = ((int64_t)b * Cout + co) * To + t;
+        if (gridDim.z == 1) {
+          float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
+          if (!act_before) o = apply_act(o, act, slope);
+          union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+          y[oi] = cv.s;
+        } else {
+          atomicAdd(&yws[oi], acc[m][n][r]);
+        }
       }
     }
+}
+
+// finalize after split-K: y = act(ws + bias)
+__global__ void deep_finalize_kernel(const float* __restrict__ yws,
+                                     const float* __restrict__ bias,
+                                     unsigned short* __restrict__ y, int B,
+                                     int Cout, int To, int act, float slope,
+                                     bool act_before) {
+  const int64_t nall = (int64_t)B * Cout * To;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nall;
+       i += step) {
+    const int co = (int)((i / To) % Cout);
+    float o = yws[i] + (bias ? bias[co] : 0.f);
+    if (!act_before) o = apply_act(o, act, slope);
+    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+    y[i] = cv.s;
+  }
 }
 
 // dot-form forward for tiny Cout (disc score conv 1024->1 k3): one block
@@ -890,14 +922,36 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
     return y;
   }
   if ((int64_t)Cin * K >= 2048 && To <= 256 && Cout >= 32) {
-    // deep contraction, small T: fold batch into 256-column tiles
-    dim3 grid(cdiv_i((int64_t)B * To, 256), cdiv_i(Cout, BM), 1);
+    // deep contraction, small T: fold batch into 256-column tiles,
+    // split-K to fill the chip when (cols x M) tiles are few
+    const int ct = cdiv_i((int64_t)B * To, 256);
+    const int mt = cdiv_i(Cout, BM);
+    const int nchunks = cdiv_i((int64_t)Cin * K, CI_CH);
+    int nsplit = (int)std::min<int64_t>(
+        nchunks, std::max<int64_t>(1, 512 / std::max(1, ct * mt)));
+    dim3 grid(ct, mt, nsplit);
+    torch::Tensor yws;
+    float* yws_p = nullptr;
+    if (nsplit > 1) {
+      yws = torch::zeros({B, Cout, (int)To},
+                         x.options().dtype(torch::kFloat));
+      yws_p = yws.data_ptr<float>();
+    }
     hipLaunchKernelGGL(conv1d_fwd_mfma_deep_kernel, grid, dim3(256), 0,
                        stream0, (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)w.data_ptr(), bias_p,
-                       (unsigned short*)y.data_ptr(), B, Cin, Tin, Cout,
-                       (int)To, K, (int)dil, (int)pad, reflect, (int)act,
-                       (float)slope, act_before);
+                       (unsigned short*)y.data_ptr(), yws_p, B, Cin, Tin,
+                       Cout, (int)To, K, (int)dil, (int)pad, reflect,
+                       (int)act, (float)slope, act_before);
+    if (nsplit > 1) {
+      const int64_t nall = (int64_t)B * Cout * To;
+      hipLaunchKernelGGL(deep_finalize_kernel,
+                         dim3((int)std::min<int64_t>(2048,
+                                                     (nall + 255) / 256)),
+                         dim3(256), 0, stream0, yws_p, bias_p,
+                         (unsigned short*)y.data_ptr(), B, Cout, (int)To,
+                         (int)act, (float)slope, act_before);
+    }
     return y;
   }
   const int span = 64 + (K - 1) * (int)dil;
