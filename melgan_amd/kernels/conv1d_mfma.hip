@@ -553,7 +553,7 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
                             float* __restrict__ yws, int B, int Cin,
                             int Tin, int Cout, int To, int K, int dil,
                             int pad, bool reflect, int act, float slope,
-                            bool act_before) {
+                            bool act_before, int l2cin, int l2to) {
   __shared__ __attribute__((aligned(16))) unsigned short smem[
       (64 + 256) * LDW];
   unsigned short* lds_w = smem;               // [64co][kk]
@@ -587,8 +587,12 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
       const int kk = idx - co * CI_CH;
       const int nn = nn0 + kk;
       unsigned short v = 0;
-      if (co0 + co < Cout && nn < NK)
-        v = w[((int64_t)(co0 + co) * Cin + nn / K) * K + nn % K];
+      if (co0 + co < Cout && nn < NK) {
+        int ci, k;
+        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
+        else { ci = nn % Cin; k = nn / Cin; }
+        v = w[((int64_t)(co0 + co) * Cin + ci) * K + k];
+      }
       lds_w[co * LDW + kk] = v;
     }
     for (int idx = tid; idx < 256 * CI_CH; idx += 256) {
@@ -598,10 +602,12 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
       unsigned short v = 0;
       const int64_t cg = cg0 + col;
       if (nn < NK && cg < NCOL) {
-        const int b = (int)(cg / To);
-        const int t = (int)(cg - (int64_t)b * To);
-        const int ci = nn / K;
-        const int k = nn - ci * K;
+        int b, t;
+        if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
+        else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
+        int ci, k;
+        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
+        else { ci = nn % Cin; k = nn / Cin; }
         const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
         v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
         if (act_before) v = leaky_bf16(v, slope);
@@ -630,8 +636,9 @@ conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
     for (int n = 0; n < 8; ++n) {
       const int64_t cg = cg0 + wc * 128 + n * 16 + col;
       if (cg >= NCOL) continue;
-      const int b = (int)(cg / To);
-      const int t = (int)(cg - (int64_t)b * To);
+      int b, t;
+      if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
+      else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int co = co0 + wr * 32 + m * 16 + row4 + r;
@@ -937,12 +944,16 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                          x.options().dtype(torch::kFloat));
       yws_p = yws.data_ptr<float>();
     }
+    auto log2i = [](int v) {
+      return (v & (v - 1)) == 0 ? __builtin_ctz(v) : -1;
+    };
     hipLaunchKernelGGL(conv1d_fwd_mfma_deep_kernel, grid, dim3(256), 0,
                        stream0, (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)w.data_ptr(), bias_p,
                        (unsigned short*)y.data_ptr(), yws_p, B, Cin, Tin,
                        Cout, (int)To, K, (int)dil, (int)pad, reflect,
-                       (int)act, (float)slope, act_before);
+                       (int)act, (float)slope, act_before, log2i(Cin),
+                       log2i((int)To));
     if (nsplit > 1) {
       const int64_t nall = (int64_t)B * Cout * To;
       hipLaunchKernelGGL(deep_finalize_kernel,
