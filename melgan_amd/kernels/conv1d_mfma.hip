@@ -928,7 +928,14 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                        (float)slope, act_before);
     return y;
   }
-  if ((int64_t)Cin * K >= 2048 && To <= 256 && Cout >= 32) {
+  static const bool use_deep = []() {
+    const char* e = getenv("MELGAN_AMD_DEEP");
+    return e && e[0] == '1';
+  }();
+  // NOTE: measured SLOWER than the specialized kernel on the disc k5
+  // shapes (51.6k vs 54.5k mel-frames/s end to end) — kept behind
+  // MELGAN_AMD_DEEP=1 for further tuning; see profiles/README.md.
+  if (use_deep && (int64_t)Cin * K >= 2048 && To <= 256 && Cout >= 32) {
     // deep contraction, small T: fold batch into 256-column tiles,
     // split-K to fill the chip when (cols x M) tiles are few
     const int ct = cdiv_i((int64_t)B * To, 256);

@@ -523,8 +523,13 @@ torch::Tensor conv1d_bwd_data(torch::Tensor gz, torch::Tensor w,
   CHECK_IN(w);
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = w.size(1) * groups, K = w.size(2);
-  if (conv1d_mfma_eligible_any(gz) && stride == 1 && groups == 1 &&
-      (int64_t)Cout * K >= 2048 && Tin + 2 * pad <= 256 && Cin >= 32) {
+  static const bool use_deep_bwd = []() {
+    const char* e = getenv("MELGAN_AMD_DEEP");
+    return e && e[0] == '1';
+  }();
+  if (use_deep_bwd && conv1d_mfma_eligible_any(gz) && stride == 1 &&
+      groups == 1 && (int64_t)Cout * K >= 2048 && Tin + 2 * pad <= 256 &&
+      Cin >= 32) {
     // deep-contraction small-T (disc k5, mel conv): flipped-weight direct
     // conv on the batch-folded 256-column kernel, then un-pad fold
     auto wf = w.permute({1, 0, 2}).flip(-1).contiguous();
