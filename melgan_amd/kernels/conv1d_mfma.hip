@@ -61,7 +61,8 @@ conv1d_fwd_mfma_kernel(const unsigned short* __restrict__ x,
                        bool reflect, int act, float slope, bool act_before) {
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   const int span = BN + (K - 1) * dil;
-  const int bufsz = (span + K * BM) * LDW;
+  unsigned short* lds_x = smem;              // [span][LDW]
+  unsigned short* lds_w = smem + span * LDW;  // [K][BM][LDW]
 
   const int t0 = blockIdx.x * BN;
   const int co0 = blockIdx.y * BM;
@@ -76,9 +77,8 @@ conv1d_fwd_mfma_kernel(const unsigned short* __restrict__ x,
   f32x4 acc[2][2] = {};
   const int gt0 = t0 - pad;
 
-  auto stage = [&](int ci0, unsigned short* buf) {
-    unsigned short* lds_x = buf;
-    unsigned short* lds_w = buf + span * LDW;
+  for (int ci0 = 0; ci0 < Cin; ci0 += CI_CH) {
+    // ---- stage x chunk: [t][ci] transposed image ----
     for (int idx = tid; idx < span * CI_CH; idx += 256) {
       const int ci = idx / span;       // consecutive tid -> consecutive t
       const int t = idx - ci * span;   // (coalesced global reads)
@@ -90,6 +90,7 @@ conv1d_fwd_mfma_kernel(const unsigned short* __restrict__ x,
       }
       lds_x[t * LDW + ci] = v;
     }
+    // ---- stage w chunk: [k][co][ci] ----
     for (int idx = tid; idx < K * BM * CI_CH; idx += 256) {
       const int k = idx / (BM * CI_CH);
       const int r = idx - k * BM * CI_CH;
@@ -100,17 +101,8 @@ conv1d_fwd_mfma_kernel(const unsigned short* __restrict__ x,
         v = w[((int64_t)(co0 + co) * Cin + ci0 + ci) * K + k];
       lds_w[(k * BM + co) * LDW + ci] = v;
     }
-  };
+    __syncthreads();
 
-  // double-buffered ci-chunk pipeline: ONE barrier per chunk; the next
-  // chunk's staging overlaps this chunk's MFMAs (guide §5.5 T3 2-phase)
-  int cur = 0;
-  stage(0, smem);
-  __syncthreads();
-  for (int ci0 = 0; ci0 < Cin; ci0 += CI_CH) {
-    if (ci0 + CI_CH < Cin) stage(ci0 + CI_CH, smem + (cur ^ 1) * bufsz);
-    const unsigned short* lds_x = smem + cur * bufsz;
-    const unsigned short* lds_w = lds_x + span * LDW;
     for (int k = 0; k < K; ++k) {
       const unsigned short* wk = lds_w + k * BM * LDW;
       const int tb = wc * 32 + row + k * dil;  // row==col for B operand
@@ -124,7 +116,6 @@ conv1d_fwd_mfma_kernel(const unsigned short* __restrict__ x,
       acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
-    cur ^= 1;
   }
 
   // ---- epilogue: bias + act, bf16 store ----
@@ -160,7 +151,8 @@ conv1d_bwd_data_mfma_kernel(const unsigned short* __restrict__ gz,
                             int dil) {
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   const int span = BN + (K - 1) * dil;
-  const int bufsz = (span + K * BM) * LDW;
+  unsigned short* lds_gz = smem;              // [t][co]
+  unsigned short* lds_w = smem + span * LDW;  // [k][ci][co]
 
   const int q0 = blockIdx.x * BN;
   const int ci0 = blockIdx.y * BM;
@@ -175,9 +167,7 @@ conv1d_bwd_data_mfma_kernel(const unsigned short* __restrict__ gz,
   f32x4 acc[2][2] = {};
   const int gt0 = q0 - (K - 1) * dil;  // gz staging origin
 
-  auto stage = [&](int co0, unsigned short* buf) {
-    unsigned short* lds_gz = buf;
-    unsigned short* lds_w = buf + span * LDW;
+  for (int co0 = 0; co0 < Cout; co0 += CI_CH) {
     for (int idx = tid; idx < span * CI_CH; idx += 256) {
       const int co = idx / span;
       const int t = idx - co * span;
@@ -197,15 +187,8 @@ conv1d_bwd_data_mfma_kernel(const unsigned short* __restrict__ gz,
         v = w[((int64_t)(co0 + co) * Cin + ci0 + ci) * K + k];
       lds_w[(k * BM + ci) * LDW + co] = v;
     }
-  };
+    __syncthreads();
 
-  int cur = 0;
-  stage(0, smem);
-  __syncthreads();
-  for (int co0 = 0; co0 < Cout; co0 += CI_CH) {
-    if (co0 + CI_CH < Cout) stage(co0 + CI_CH, smem + (cur ^ 1) * bufsz);
-    const unsigned short* lds_gz = smem + cur * bufsz;
-    const unsigned short* lds_w = lds_gz + span * LDW;
     for (int k = 0; k < K; ++k) {
       const unsigned short* wk = lds_w + k * BM * LDW;
       // q - k*d  ->  staged t = q_local + (K-1-k)*d
@@ -220,7 +203,6 @@ conv1d_bwd_data_mfma_kernel(const unsigned short* __restrict__ gz,
       acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
-    cur ^= 1;
   }
 
   const int col = lane & 15;
@@ -769,7 +751,9 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
                               int dil, int pad, bool reflect, int groups,
                               float slope, bool act_a, bool act_b,
                               int nsplit) {
-  __shared__ __attribute__((aligned(16))) unsigned short smem[2][2 * 64 * LDT];
+  __shared__ __attribute__((aligned(16))) unsigned short smem[2 * 64 * LDT];
+  unsigned short* lds_gz = smem;            // [co][t]
+  unsigned short* lds_xs = smem + 64 * LDT;  // [col=(cig,k)][t]
 
   const int Copg = Cout / groups;
   const int Cpg = Cin / groups;
@@ -791,11 +775,11 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
 
   f32x4 acc[2][2] = {};
 
-  auto stage = [&](int64_t c, unsigned short* buf) {
-    unsigned short* lds_gz = buf;
-    unsigned short* lds_xs = buf + 64 * LDT;
+  for (int64_t c = blockIdx.z; c < total; c += nsplit) {
     const int b = (int)(c / tchunks);
     const int t0 = (int)(c - (int64_t)b * tchunks) * TCH;
+    // A stage (gz rows, always t-contiguous): ushort4 when the whole
+    // 64-t run is interior, scalar tail otherwise
     const bool a_vec = (t0 + TCH <= To) && !act_a;
     if (a_vec) {
       for (int idx = tid; idx < 64 * (TCH / 4); idx += 256) {
@@ -819,6 +803,8 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
         lds_gz[co * LDT + t] = v;
       }
     }
+    // B stage (shifted x rows): vectorize per-column when the run is
+    // interior, 8-byte aligned and stride-1
     for (int idx = tid; idx < 64 * (TCH / 4); idx += 256) {
       const int cc = idx / (TCH / 4);
       const int t4 = (idx - cc * (TCH / 4)) * 4;
@@ -853,18 +839,8 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
       }
       *(ushort4*)(lds_xs + cc * LDT + t4) = v;
     }
-  };
+    __syncthreads();
 
-  // double-buffered chunk pipeline: ONE barrier per chunk; the next
-  // chunk's staging overlaps this chunk's MFMAs (guide §5.5 T3 minimum
-  // 2-phase form)
-  int cur = 0;
-  if (blockIdx.z < total) stage(blockIdx.z, smem[0]);
-  __syncthreads();
-  for (int64_t c = blockIdx.z; c < total; c += nsplit) {
-    if (c + nsplit < total) stage(c + nsplit, smem[cur ^ 1]);
-    const unsigned short* lds_gz = smem[cur];
-    const unsigned short* lds_xs = smem[cur] + 64 * LDT;
 #pragma unroll
     for (int step = 0; step < 2; ++step) {
       const int tt = step * 32 + (lane >> 4) * 8;
@@ -878,7 +854,6 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
       acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
-    cur ^= 1;
   }
 
   const int col = lane & 15;
@@ -998,18 +973,7 @@ torch::Tensor conv1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
     return y;
   }
   const int span = 64 + (K - 1) * (int)dil;
-  const size_t lds =
-      2 * (size_t)(span + K * 64) * LDW * sizeof(unsigned short);
-  static bool attr_set = []() {
-    hipFuncSetAttribute((const void*)conv1d_fwd_mfma_kernel,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        144 * 1024);
-    hipFuncSetAttribute((const void*)conv1d_bwd_data_mfma_kernel,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        144 * 1024);
-    return true;
-  }();
-  (void)attr_set;
+  const size_t lds = (size_t)(span + K * 64) * LDW * sizeof(unsigned short);
   dim3 grid(cdiv_i(To, BN), cdiv_i(Cout, BM), B);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv1d_fwd_mfma_kernel, grid, dim3(256), lds, stream,
@@ -1075,15 +1039,7 @@ torch::Tensor conv1d_bwd_data_mfma(torch::Tensor gz, torch::Tensor w,
   const int Tpad = (int)Tin + 2 * (int)pad;
   auto gxpad = torch::empty({B, Cin, Tpad}, gz.options());
   const int span = 64 + (K - 1) * (int)dil;
-  const size_t lds =
-      2 * (size_t)(span + K * 64) * LDW * sizeof(unsigned short);
-  static bool attr_set2 = []() {
-    hipFuncSetAttribute((const void*)conv1d_bwd_data_mfma_kernel,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        144 * 1024);
-    return true;
-  }();
-  (void)attr_set2;
+  const size_t lds = (size_t)(span + K * 64) * LDW * sizeof(unsigned short);
   dim3 grid(cdiv_i(Tpad, BN), cdiv_i(Cin, BM), B);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv1d_bwd_data_mfma_kernel, grid, dim3(256), lds,
