@@ -743,6 +743,107 @@ __global__ void zero_stuff_kernel(const unsigned short* __restrict__ x,
 constexpr int TCH = 64;   // t per chunk
 constexpr int LDT = 72;   // padded t stride (144 B rows)
 
+// 16-row weight-grad variant for small groups (disc k=41 towers,
+// Copg in {4,16}): cuts the 64-row tile's masked-MFMA waste 4-16x.
+// waves: 4, each owns 16 of the 64 (cig,k) columns; M tile = 16 co.
+// grid: (ceil(Cpg*K/64), G*ceil(Copg/16), NSPLIT)
+__global__ void __launch_bounds__(256)
+conv1d_bwd_weight_mfma16_kernel(const unsigned short* __restrict__ gz,
+                                const unsigned short* __restrict__ x,
+                                float* __restrict__ gw, int B, int Cin,
+                                int Tin, int Cout, int To, int K,
+                                int stride, int dil, int pad, bool reflect,
+                                int groups, float slope, bool act_a,
+                                bool act_b, int nsplit) {
+  __shared__ __attribute__((aligned(16))) unsigned short smem[
+      (16 + 64) * LDT];
+  unsigned short* lds_gz = smem;             // [16co][t]
+  unsigned short* lds_xs = smem + 16 * LDT;  // [col=(cig,k)][t]
+
+  const int Copg = Cout / groups;
+  const int Cpg = Cin / groups;
+  const int mtiles = (Copg + 15) / 16;
+  const int g = blockIdx.y / mtiles;
+  const int mt = blockIdx.y - g * mtiles;
+  const int co_base = g * Copg + mt * 16;
+  const int co_end = (g + 1) * Copg;
+  const int n0 = blockIdx.x * 64;
+  const int N = Cpg * K;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // wave owns cols [wid*16, wid*16+16)
+  const int row = lane & 15;
+
+  const int tchunks = (To + TCH - 1) / TCH;
+  const int64_t total = (int64_t)B * tchunks;
+
+  f32x4 acc = {};
+
+  for (int64_t c = blockIdx.z; c < total; c += nsplit) {
+    const int b = (int)(c / tchunks);
+    const int t0 = (int)(c - (int64_t)b * tchunks) * TCH;
+    for (int idx = tid; idx < 16 * TCH; idx += 256) {
+      const int co = idx / TCH;
+      const int t = idx - co * TCH;
+      unsigned short v = 0;
+      if (co_base + co < co_end && t0 + t < To) {
+        v = gz[((int64_t)b * Cout + co_base + co) * To + t0 + t];
+        if (act_a) v = leaky_bf16(v, slope);
+      }
+      lds_gz[co * LDT + t] = v;
+    }
+    for (int idx = tid; idx < 64 * (TCH / 4); idx += 256) {
+      const int cc = idx / (TCH / 4);
+      const int t4 = (idx - cc * (TCH / 4)) * 4;
+      const int nn = n0 + cc;
+      ushort4 v = {0, 0, 0, 0};
+      if (nn < N) {
+        const int ci = g * Cpg + nn / K;
+        const int k = nn % K;
+        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          unsigned short e = 0;
+          if (t0 + t4 + u < To) {
+            e = load_pad_bf16(xr, (t0 + t4 + u) * stride + k * dil - pad,
+                              Tin, reflect);
+            if (act_b) e = leaky_bf16(e, slope);
+          }
+          (&v.x)[u] = e;
+        }
+      }
+      *(ushort4*)(lds_xs + cc * LDT + t4) = v;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int step = 0; step < 2; ++step) {
+      const int tt = step * 32 + (lane >> 4) * 8;
+      bf16x8 a0 = *(const bf16x8*)(lds_gz + row * LDT + tt);
+      bf16x8 b0 = *(const bf16x8*)(lds_xs + (wid * 16 + row) * LDT + tt);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+  {
+    const int nn = n0 + wid * 16 + col;
+    if (nn < N) {
+      const int cig = nn / K;
+      const int k = nn - cig * K;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co_base + row4 + r;
+        if (co < co_end && co < Cout)
+          atomicAdd(&gw[((int64_t)co * Cpg + cig) * K + k], acc[r]);
+      }
+    }
+  }
+}
+
+
 __global__ void __launch_bounds__(256)
 conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
                               const unsigned short* __restrict__ x,
@@ -1127,6 +1228,23 @@ torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
   const int N = Cpg * (int)K;
   auto gw = torch::zeros({Cout, Cpg, (int)K},
                          x.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  if (Copg <= 16) {
+    const int gx16 = cdiv_i(N, 64);
+    const int gy16 = (int)groups * cdiv_i(Copg, 16);
+    const int tchunks16 = cdiv_i(To, TCH);
+    const int64_t total16 = (int64_t)B * tchunks16;
+    int ns16 = (int)std::min<int64_t>(
+        total16, std::max<int64_t>(1, 2048 / std::max(1, gx16 * gy16)));
+    dim3 grid16(gx16, gy16, ns16);
+    hipLaunchKernelGGL(conv1d_bwd_weight_mfma16_kernel, grid16, dim3(256),
+                       0, stream, (const unsigned short*)gz.data_ptr(),
+                       (const unsigned short*)x.data_ptr(),
+                       gw.data_ptr<float>(), B, Cin, Tin, Cout, To, (int)K,
+                       (int)stride, (int)dil, (int)pad, reflect,
+                       (int)groups, (float)slope, act_a, act_b, ns16);
+    return gw;
+  }
   const int gx_ = cdiv_i(N, 64);
   const int gy_ = (int)groups * cdiv_i(Copg, 64);
   const int tchunks = cdiv_i(To, TCH);
@@ -1134,7 +1252,6 @@ torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
   int nsplit = (int)std::min<int64_t>(
       total, std::max<int64_t>(1, 2048 / std::max(1, gx_ * gy_)));
   dim3 grid(gx_, gy_, nsplit);
-  auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv1d_bwd_weight_mfma_kernel, grid, dim3(256), 0,
                      stream, (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
