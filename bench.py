@@ -143,28 +143,34 @@ def main():
         print(json.dumps(out))
 
 
-def measure_rtf(hp, device, n_frames: int = 860, reps: int = 20):
-    """Inference RTF @22.05 kHz: ~10 s of audio per run, fused fwd path."""
+def measure_rtf(hp, device, n_frames: int = 860, reps: int = 20,
+                dtype=torch.bfloat16):
+    """Inference RTF: ~10 s of audio per run on the fused MFMA fwd path
+    (bf16 weights/activations; PQMF synthesis fp32)."""
     from melgan_amd.models import build_generator
     from melgan_amd.ops.pqmf import PQMF
 
     g = build_generator(hp).to(device).eval()
     g.remove_weight_norm()
+    if device.type == "cuda" and dtype is not None:
+        g = g.to(dtype)
     pqmf = PQMF(hp.model.out_channels).to(device) \
         if hp.model.out_channels > 1 else None
     mel = (torch.randn(1, hp.audio.n_mel_channels, n_frames,
                        device=device) * 2.0 - 6.0)
+    if device.type == "cuda" and dtype is not None:
+        mel = mel.to(dtype)
     sr = hp.audio.sampling_rate
     with torch.no_grad():
         for _ in range(3):
             sub = g(mel)
-            audio = pqmf.synthesis(sub) if pqmf else sub
+            audio = pqmf.synthesis(sub.float()) if pqmf else sub
         if device.type == "cuda":
             torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(reps):
             sub = g(mel)
-            audio = pqmf.synthesis(sub) if pqmf else sub
+            audio = pqmf.synthesis(sub.float()) if pqmf else sub
         if device.type == "cuda":
             torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / reps

@@ -46,6 +46,8 @@ def main():
                     help="folder of mel .npy/.mel files")
     ap.add_argument("-o", "--output_folder", default="generated")
     ap.add_argument("-d", "--device", default=None)
+    ap.add_argument("--fp32", action="store_true",
+                    help="run the generator in fp32 (default bf16 on GPU)")
     args = ap.parse_args()
 
     hp = merge_default(HParam(args.config))
@@ -59,6 +61,9 @@ def main():
     load_flexible(generator, state)
     generator.eval()
     generator.remove_weight_norm()
+    use_bf16 = device.type == "cuda" and not args.fp32
+    if use_bf16:
+        generator = generator.to(torch.bfloat16)
     multiband = hp.model.out_channels > 1
     pqmf = PQMF(hp.model.out_channels,
                 cutoff_ratio=hp.model.get("pqmf_cutoff")).to(device) \
@@ -83,10 +88,12 @@ def main():
             if mel.dim() == 2:
                 mel = mel.unsqueeze(0)
             mel = mel.to(device)
+            if use_bf16:
+                mel = mel.to(torch.bfloat16)
             if device.type == "cuda":
                 torch.cuda.synchronize()
             t0 = time.perf_counter()
-            audio_sub = generator.inference(mel)
+            audio_sub = generator.inference(mel).float()
             audio = pqmf.synthesis(audio_sub) if multiband else audio_sub
             if device.type == "cuda":
                 torch.cuda.synchronize()

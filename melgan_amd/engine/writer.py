@@ -74,6 +74,9 @@ class MetricsWriter:
                 try:
                     self.tb.add_audio("val/audio", audio.reshape(1, -1),
                                       step, sample_rate=sr)
+                    fig = _spec_figure(audio)
+                    if fig is not None:
+                        self.tb.add_figure("val/spectrogram", fig, step)
                 except Exception:
                     pass
 
@@ -82,3 +85,25 @@ class MetricsWriter:
             self.jsonl.close()
             if self.tb is not None:
                 self.tb.close()
+
+
+def _spec_figure(audio, n_fft: int = 1024, hop: int = 256):
+    """Matplotlib spectrogram figure for TensorBoard (family parity)."""
+    try:
+        import matplotlib
+        matplotlib.use("Agg")
+        import matplotlib.pyplot as plt
+        import torch
+
+        spec = torch.stft(audio.reshape(1, -1).float(), n_fft, hop,
+                          window=torch.hann_window(n_fft),
+                          return_complex=True).abs()[0]
+        fig, ax = plt.subplots(figsize=(8, 3))
+        ax.imshow(spec.clamp_min(1e-5).log().numpy(), origin="lower",
+                  aspect="auto", cmap="magma")
+        ax.set_xlabel("frame")
+        ax.set_ylabel("bin")
+        fig.tight_layout()
+        return fig
+    except Exception:
+        return None

@@ -2,18 +2,18 @@
 # MFMA tier validation + bench + profile.
 set -x
 cd /root/repo
-mkdir -p gpurun_out/prof13
+mkdir -p gpurun_out/prof14
 timeout 900 python -m pytest tests/test_kernels_gpu.py -q -m gpu --timeout 300 \
-    > gpurun_out/gpu_tests13.log 2>&1
+    > gpurun_out/gpu_tests14.log 2>&1
 echo "PYTEST EXIT: $?"
-tail -15 gpurun_out/gpu_tests13.log
+tail -15 gpurun_out/gpu_tests14.log
 timeout 600 python bench.py --steps 15 --warmup 5 --inference-rtf \
-    > gpurun_out/bench_mfma12.log 2>&1
+    > gpurun_out/bench_final1.log 2>&1
 echo "BENCH EXIT: $?"
-tail -2 gpurun_out/bench_mfma12.log
+tail -2 gpurun_out/bench_final1.log
 export TMPDIR=/tmp
 cd /tmp
-timeout 600 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof13 -o r13 -- \
+timeout 600 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof14 -o r14 -- \
     python /root/repo/bench.py --steps 5 --warmup 2 --batch-size 32 \
-    > /root/repo/gpurun_out/prof13_bench.log 2>&1
+    > /root/repo/gpurun_out/prof14_bench.log 2>&1
 echo "PROF EXIT: $?"

@@ -558,3 +558,21 @@ def test_stft_loss_module_gpu():
     sc_c, mag_c = loss_cpu(x, y)
     assert abs(float(sc_g) - float(sc_c)) < 1e-3
     assert abs(float(mag_g) - float(mag_c)) < 1e-3
+
+
+def test_bf16_inference_path():
+    """remove_weight_norm + bf16 generator forward (the RTF path)."""
+    from melgan_amd.models.generator import Generator
+
+    torch.manual_seed(0)
+    g = Generator(80, 4, 256, [8, 4, 2], 3).to(DEV).eval()
+    mel = torch.randn(1, 80, 64, device=DEV)
+    with torch.no_grad():
+        y32 = g(mel)
+        g.remove_weight_norm()
+        y32b = g(mel)
+        g16 = g.to(torch.bfloat16)
+        y16 = g16(mel.to(torch.bfloat16))
+    assert torch.allclose(y32, y32b, atol=1e-3)
+    rel = (y16.float() - y32).abs().mean() / y32.abs().mean()
+    assert rel < 5e-2, f"bf16 inference rel err {float(rel):.4f}"
