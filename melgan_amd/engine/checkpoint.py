@@ -57,26 +57,49 @@ def load_flexible(module: torch.nn.Module, state: dict, strict_first=True,
             logger.info("strict checkpoint load failed; trying ordered "
                         "shape-matched load")
     own = module.state_dict()
-    own_items = [(k, v) for k, v in own.items()]
-    src_items = [(k, v) for k, v in state.items()
-                 if isinstance(v, torch.Tensor)]
 
     def kind(key: str) -> str:
         return key.rsplit(".", 1)[-1]
 
+    def prefix(key: str) -> str:
+        return key.rsplit(".", 1)[0] if "." in key else ""
+
+    def modules_in_order(items):
+        """[(prefix, {kind: (key, tensor)})] preserving insertion order.
+        A module = the group of tensors sharing a key prefix (one conv's
+        weight_v/weight_g/bias, in whatever per-module key ORDER the
+        source library emits them)."""
+        groups = []
+        index = {}
+        for k, v in items:
+            p = prefix(k)
+            if p not in index:
+                index[p] = len(groups)
+                groups.append((p, {}))
+            groups[index[p]][1][kind(k)] = (k, v)
+        return groups
+
+    src_groups = modules_in_order(
+        [(k, v) for k, v in state.items() if isinstance(v, torch.Tensor)])
+    own_groups = modules_in_order(list(own.items()))
+
     mapped = {}
     si = 0
-    for ok, ov in own_items:
+    for op, okinds in own_groups:
         found = False
-        for j in range(si, len(src_items)):
-            sk, sv = src_items[j]
-            if kind(sk) == kind(ok) and tuple(sv.shape) == tuple(ov.shape):
-                mapped[ok] = sv
+        for j in range(si, len(src_groups)):
+            sp, skinds = src_groups[j]
+            if set(okinds) == set(skinds) and all(
+                    tuple(skinds[t][1].shape) == tuple(okinds[t][1].shape)
+                    for t in okinds):
+                for t in okinds:
+                    mapped[okinds[t][0]] = skinds[t][1]
                 si = j + 1
                 found = True
                 break
         if not found:
+            shapes = {t: tuple(v.shape) for t, (_, v) in okinds.items()}
             raise RuntimeError(
-                f"checkpoint translation failed: no source tensor for "
-                f"{ok} (shape {tuple(ov.shape)}, kind {kind(ok)})")
+                f"checkpoint translation failed: no source module for "
+                f"'{op}' with tensors {shapes}")
     module.load_state_dict(mapped)

@@ -58,6 +58,13 @@ class Trainer:
                 torch.cuda.device_count(), 1)) \
                 if torch.cuda.is_available() else torch.device("cpu")
         self.device = device
+        if hp.backend.get("debug_sync"):
+            # deterministic launch-order debugging:
+            # serialise kernel launches and make HIP errors synchronous
+            import os as _os
+
+            _os.environ.setdefault("HIP_LAUNCH_BLOCKING", "1")
+            _os.environ.setdefault("AMD_SERIALIZE_KERNEL", "3")
 
         self.generator = build_generator(hp).to(device)
         self.discriminator = build_discriminator(hp).to(device)
