@@ -1,0 +1,108 @@
+"""End-to-end CLI round trip on CPU (reference workflow parity):
+preprocess.py (wav -> .mel) -> train.py (few steps, ckpt) ->
+inference.py (.mel/.npy -> .wav, RTF report)."""
+
+import json
+import os
+import subprocess
+import sys
+import wave
+
+import numpy as np
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+TINY_CFG = """
+data: {{train_dir: {d}, val_dir: {d}}}
+train:
+  batch_size: 2
+  num_workers: 0
+  discriminator_train_start_steps: 0
+audio:
+  n_mel_channels: 80
+  segment_length: 4096
+  pad_short: 2000
+  hop_length: 256
+  filter_length: 1024
+  win_length: 1024
+  sampling_rate: 22050
+model: {{out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}}
+log: {{summary_interval: 2, validation_interval: 1000, save_interval: 3,
+      chkpt_dir: {ck}, log_dir: {lg}}}
+backend: {{amp: none, fused_adam: false}}
+"""
+
+
+def _write_wav(path, seconds=1.0, sr=22050):
+    t = np.arange(int(seconds * sr))
+    sig = (0.4 * np.sin(2 * np.pi * 220 * t / sr) +
+           0.1 * np.random.RandomState(0).randn(len(t)))
+    pcm = np.clip(sig * 32767, -32768, 32767).astype(np.int16)
+    with wave.open(path, "wb") as w:
+        w.setnchannels(1)
+        w.setsampwidth(2)
+        w.setframerate(sr)
+        w.writeframes(pcm.tobytes())
+
+
+def _run(args, **kw):
+    return subprocess.run([sys.executable] + args, cwd=REPO,
+                          capture_output=True, text=True, timeout=600, **kw)
+
+
+@pytest.mark.timeout(900)
+def test_preprocess_train_infer_roundtrip(tmp_path):
+    data = tmp_path / "wavs"
+    data.mkdir()
+    for i in range(3):
+        _write_wav(str(data / f"utt{i}.wav"), seconds=0.5)
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text(TINY_CFG.format(d=str(data), ck=str(tmp_path / "ck"),
+                                   lg=str(tmp_path / "lg")))
+
+    # preprocess: wav -> .mel
+    r = _run(["preprocess.py", "-c", str(cfg), "-d", str(data)])
+    assert r.returncode == 0, r.stderr[-2000:]
+    mels = list(data.glob("*.mel"))
+    assert len(mels) == 3
+
+    # train 3 steps on the real (tiny) dataset; checkpoint at step 3
+    r = _run(["train.py", "-c", str(cfg), "-n", "e2e", "--max-steps", "3"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    cks = sorted((tmp_path / "ck").glob("e2e_*.pt"))
+    assert cks, "no checkpoint written"
+
+    # inference: mel .npy -> wav with RTF report
+    meldir = tmp_path / "mels"
+    meldir.mkdir()
+    mel = torch.load(mels[0], weights_only=True).numpy()
+    np.save(str(meldir / "utt0.npy"), mel)
+    out = tmp_path / "wav_out"
+    r = _run(["inference.py", "-c", str(cfg), "-p", str(cks[-1]),
+              "-i", str(meldir), "-o", str(out)])
+    assert r.returncode == 0, r.stderr[-2000:]
+    wavs = list(out.glob("*.wav"))
+    assert len(wavs) == 1
+    assert "TOTAL RTF" in r.stdout
+    with wave.open(str(wavs[0]), "rb") as w:
+        assert w.getframerate() == 22050
+        assert w.getnframes() > 10000
+
+
+@pytest.mark.timeout(300)
+def test_bench_cpu_contract(tmp_path):
+    """bench.py emits the driver-contract JSON line (CPU, tiny)."""
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text(TINY_CFG.format(d="null", ck=str(tmp_path / "ck"),
+                                   lg=str(tmp_path / "lg")))
+    r = _run(["bench.py", "--steps", "2", "--warmup", "1",
+              "--config", str(cfg), "--batch-size", "2"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    rec = json.loads(line)
+    assert rec["metric"] == "train_mel_frames_per_sec"
+    assert rec["n_gpus"] == 1 and rec["steps"] == 2
+    assert rec["value"] > 0
+    assert rec["config"]["parallelism"] == "dp1"
