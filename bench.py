@@ -106,6 +106,9 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t)
 
+    peak_hbm_gb = round(torch.cuda.max_memory_allocated() / 2 ** 30, 2) \
+        if have_gpu else None
+
     rtf = None
     if args.inference_rtf and rank == 0:
         rtf = measure_rtf(hp, device)
@@ -135,6 +138,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "segment_samples": seg,
                 "adversarial": args.adversarial,
+                "peak_hbm_gb": peak_hbm_gb,
             },
         }
         if rtf is not None:
