@@ -35,11 +35,6 @@ def load_checkpoint(path: str, map_location="cpu") -> dict:
     return torch.load(path, map_location=map_location, weights_only=False)
 
 
-def _order_key(name: str):
-    """Sort key preserving module order from state-dict insertion order."""
-    return 0
-
-
 def load_flexible(module: torch.nn.Module, state: dict, strict_first=True,
                   logger=None) -> None:
     """Load `state` into `module`; fall back to ordered shape matching.

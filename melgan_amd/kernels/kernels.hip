@@ -10,8 +10,9 @@
 //   - wave64: blocks are multiples of 64; the t (time) axis maps to lanes
 //     so global loads coalesce; the co axis maps to waves so weight reads
 //     are wave-uniform (scalar-cache path).
-//   - weight-norm is folded on the PYTHON side (w = g*v/||v|| computed
-//     once per step); kernels see plain weights (K5 eliminated).
+//   - weight-norm is folded by the wn_fold kernel (w = g*v/||v||, one
+//     launch, cached per optimizer step); conv kernels see plain
+//     weights (K5 eliminated from the hot path).
 //   - reflection/zero padding is folded into the address calculation
 //     (K8 eliminated); the reflect *backward* edge-fold is computed
 //     in-place by evaluating the mirrored padded positions.

@@ -23,8 +23,6 @@ import torch.nn as nn
 
 from melgan_amd.ops.wnconv import WNConv1d, WNConvTranspose1d
 
-MAX_WAV_VALUE = 32768.0
-
 
 class ResBlock(nn.Module):
     """One dilated residual block: x + f(x),
