@@ -476,6 +476,11 @@ torch::Tensor stft_frame_bwd(torch::Tensor gframes, torch::Tensor window,
                              int64_t S);
 torch::Tensor complex_mag_clamp(torch::Tensor spec_ri, double mn);
 torch::Tensor mag_bwd(torch::Tensor gmag, torch::Tensor spec_ri, double mn);
+torch::Tensor avg_pool1d_421(torch::Tensor x);
+torch::Tensor avg_pool1d_421_bwd(torch::Tensor gy, int64_t Tin);
+torch::Tensor stft_loss_reduce(torch::Tensor mx, torch::Tensor my);
+torch::Tensor stft_loss_bwd(torch::Tensor mx, torch::Tensor my,
+                            torch::Tensor sums, torch::Tensor gouts);
 
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
@@ -863,6 +868,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stft_frame_bwd", &stft_frame_bwd);
   m.def("complex_mag_clamp", &complex_mag_clamp);
   m.def("mag_bwd", &mag_bwd);
+  m.def("avg_pool1d_421", &avg_pool1d_421);
+  m.def("avg_pool1d_421_bwd", &avg_pool1d_421_bwd);
+  m.def("stft_loss_reduce", &stft_loss_reduce);
+  m.def("stft_loss_bwd", &stft_loss_bwd);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

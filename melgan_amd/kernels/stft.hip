@@ -127,6 +127,126 @@ __global__ void unpad_fold_stft_kernel(const T* __restrict__ gxpad,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K4: discriminator input down-scaling  avg_pool1d(k=4, s=2, pad=1,
+// count_include_pad=False)
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void avg_pool1d_kernel(const T* __restrict__ x,
+                                  T* __restrict__ y, int64_t BC, int Tin,
+                                  int To) {
+  const int64_t n = BC * To;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const int64_t bc = i / To;
+    const int t = (int)(i - bc * To);
+    const T* row = x + bc * Tin;
+    float acc = 0.f;
+    int cnt = 0;
+    const int base = t * 2 - 1;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int q = base + j;
+      if (q >= 0 && q < Tin) {
+        acc += static_cast<float>(row[q]);
+        ++cnt;
+      }
+    }
+    y[i] = static_cast<T>(acc / cnt);
+  }
+}
+
+template <typename T>
+__global__ void avg_pool1d_bwd_kernel(const T* __restrict__ gy,
+                                      T* __restrict__ gx, int64_t BC,
+                                      int Tin, int To) {
+  const int64_t n = BC * Tin;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const int64_t bc = i / Tin;
+    const int q = (int)(i - bc * Tin);
+    const T* gr = gy + bc * To;
+    float acc = 0.f;
+    // outputs t whose window [2t-1, 2t+2] contains q
+    for (int t = max(0, (q - 2 + 1) / 2); t <= (q + 1) / 2 && t < To; ++t) {
+      const int base = t * 2 - 1;
+      if (q < base || q > base + 3) continue;
+      int cnt = 0;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int qq = base + j;
+        if (qq >= 0 && qq < Tin) ++cnt;
+      }
+      acc += static_cast<float>(gr[t]) / cnt;
+    }
+    gx[i] = static_cast<T>(acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K11: fused STFT-loss reductions over a pair of magnitude tensors:
+//   out[0] = sum((my - mx)^2)   (spectral-convergence numerator^2)
+//   out[1] = sum(my^2)          (denominator^2)
+//   out[2] = sum(|log mx - log my|)
+// ---------------------------------------------------------------------------
+__global__ void stft_loss_reduce_kernel(const float* __restrict__ mx,
+                                        const float* __restrict__ my,
+                                        float* __restrict__ out, int64_t n) {
+  __shared__ float red[3][256];
+  float a = 0.f, b = 0.f, c = 0.f;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const float x = mx[i], y = my[i];
+    const float d = y - x;
+    a += d * d;
+    b += y * y;
+    c += fabsf(logf(x) - logf(y));
+  }
+  red[0][threadIdx.x] = a;
+  red[1][threadIdx.x] = b;
+  red[2][threadIdx.x] = c;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off)
+      for (int r = 0; r < 3; ++r)
+        red[r][threadIdx.x] += red[r][threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(&out[0], red[0][0]);
+    atomicAdd(&out[1], red[1][0]);
+    atomicAdd(&out[2], red[2][0]);
+  }
+}
+
+// backward wrt mx only (my is the no-grad target):
+//   d sc/d mx = (mx - my) / (sqrt(num)*sqrt(den))
+//   d l1/d mx = sign(log mx - log my) / (n * mx)
+// gx = gsc * dsc + gl1 * dl1
+__global__ void stft_loss_bwd_kernel(const float* __restrict__ mx,
+                                     const float* __restrict__ my,
+                                     const float* __restrict__ sums,
+                                     const float* __restrict__ gouts,
+                                     float* __restrict__ gx, int64_t n) {
+  const float gsc = gouts[0], gl1 = gouts[1];
+  const float num = sqrtf(sums[0]);
+  const float den = fmaxf(sqrtf(sums[1]), 1e-7f);
+  const float inv_scden = 1.f / fmaxf(num * den, 1e-12f);
+  const float inv_n = 1.f / (float)n;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += step) {
+    const float x = mx[i], y = my[i];
+    float g = gsc * (x - y) * inv_scden;
+    const float ld = logf(x) - logf(y);
+    g += gl1 * ((ld > 0.f) ? 1.f : (ld < 0.f ? -1.f : 0.f)) * inv_n / x;
+    gx[i] = g;
+  }
+}
+
 }  // namespace melgan
 
 using namespace melgan;
@@ -199,4 +319,60 @@ torch::Tensor mag_bwd(torch::Tensor gmag, torch::Tensor spec_ri, double mn) {
                      gmag.data_ptr<float>(), spec_ri.data_ptr<float>(),
                      gspec.data_ptr<float>(), n, (float)mn);
   return gspec;
+}
+
+torch::Tensor avg_pool1d_421(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int64_t BC = x.size(0) * x.size(1);
+  const int Tin = x.size(2);
+  const int To = (Tin + 2 * 1 - 4) / 2 + 1;
+  auto y = torch::empty({x.size(0), x.size(1), To}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, x.scalar_type(), "avgpool", [&] {
+        hipLaunchKernelGGL((avg_pool1d_kernel<scalar_t>),
+                           dim3(gblocks(BC * To)), dim3(256), 0, stream,
+                           x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           BC, Tin, To);
+      });
+  return y;
+}
+
+torch::Tensor avg_pool1d_421_bwd(torch::Tensor gy, int64_t Tin) {
+  TORCH_CHECK(gy.is_cuda() && gy.is_contiguous());
+  const int64_t BC = gy.size(0) * gy.size(1);
+  const int To = gy.size(2);
+  auto gx = torch::empty({gy.size(0), gy.size(1), (int)Tin}, gy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, gy.scalar_type(), "avgpool_bwd", [&] {
+        hipLaunchKernelGGL((avg_pool1d_bwd_kernel<scalar_t>),
+                           dim3(gblocks(BC * Tin)), dim3(256), 0, stream,
+                           gy.data_ptr<scalar_t>(), gx.data_ptr<scalar_t>(),
+                           BC, (int)Tin, To);
+      });
+  return gx;
+}
+
+torch::Tensor stft_loss_reduce(torch::Tensor mx, torch::Tensor my) {
+  TORCH_CHECK(mx.is_cuda() && mx.is_contiguous() && my.is_contiguous());
+  auto out = torch::zeros({3}, mx.options());
+  const int64_t n = mx.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(stft_loss_reduce_kernel, dim3(gblocks(n)), dim3(256),
+                     0, stream, mx.data_ptr<float>(), my.data_ptr<float>(),
+                     out.data_ptr<float>(), n);
+  return out;
+}
+
+torch::Tensor stft_loss_bwd(torch::Tensor mx, torch::Tensor my,
+                            torch::Tensor sums, torch::Tensor gouts) {
+  auto gx = torch::empty_like(mx);
+  const int64_t n = mx.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(stft_loss_bwd_kernel, dim3(gblocks(n)), dim3(256), 0,
+                     stream, mx.data_ptr<float>(), my.data_ptr<float>(),
+                     sums.data_ptr<float>(), gouts.data_ptr<float>(),
+                     gx.data_ptr<float>(), n);
+  return gx;
 }

@@ -66,11 +66,20 @@ class MultiScaleDiscriminator(nn.Module):
 
     def forward(self, audio: torch.Tensor):
         """audio: (B, 1, S) -> list over scales of [feat..., score]."""
+        from melgan_amd.ops import _ext
+
         results = []
         x = audio
+        use_hip = _ext.use_hip(audio) and _ext.ext() is not None and \
+            hasattr(_ext.ext(), "avg_pool1d_421")
         for i, tower in enumerate(self.towers):
             if i > 0:
-                x = F.avg_pool1d(x, kernel_size=4, stride=2, padding=1,
-                                 count_include_pad=False)
+                if use_hip:
+                    from melgan_amd.ops.fused import fused_avg_pool421
+
+                    x = fused_avg_pool421(x)
+                else:
+                    x = F.avg_pool1d(x, kernel_size=4, stride=2, padding=1,
+                                     count_include_pad=False)
             results.append(tower(x))
         return results

@@ -280,3 +280,50 @@ class _MagClampFn(torch.autograd.Function):
         (spec_ri,) = ctx.saved_tensors
         gspec_ri = C.mag_bwd(gmag.contiguous(), spec_ri, 1e-7)
         return torch.view_as_complex(gspec_ri)
+
+
+class _STFTPairLossFn(torch.autograd.Function):
+    """Fused SC + log-mag loss over one resolution (K11): one reduction
+    kernel forward, one elementwise kernel backward; grads stay on
+    device (no host sync)."""
+
+    @staticmethod
+    def forward(ctx, mx, my):
+        C = _ext.require_ext()
+        sums = C.stft_loss_reduce(mx, my)  # [num^2, den^2, l1_sum]
+        sc = sums[0].sqrt() / sums[1].sqrt().clamp_min(1e-7)
+        mag = sums[2] / mx.numel()
+        ctx.save_for_backward(mx, my, sums)
+        return sc, mag
+
+    @staticmethod
+    def backward(ctx, gsc, gmag):
+        C = _ext.require_ext()
+        mx, my, sums = ctx.saved_tensors
+        gouts = torch.stack([gsc.detach().float(),
+                             gmag.detach().float()]).contiguous()
+        gx = C.stft_loss_bwd(mx, my, sums, gouts)
+        return gx, None
+
+
+def fused_stft_pair_loss(mx, my):
+    return _STFTPairLossFn.apply(mx.contiguous(), my.contiguous())
+
+
+class _AvgPool421Fn(torch.autograd.Function):
+    """avg_pool1d(k=4, s=2, p=1, count_include_pad=False) — K4."""
+
+    @staticmethod
+    def forward(ctx, x):
+        C = _ext.require_ext()
+        ctx.tin = x.size(2)
+        return C.avg_pool1d_421(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        return C.avg_pool1d_421_bwd(gy.contiguous(), ctx.tin)
+
+
+def fused_avg_pool421(x):
+    return _AvgPool421Fn.apply(x)

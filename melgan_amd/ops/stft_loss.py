@@ -66,6 +66,11 @@ class STFTLoss(nn.Module):
                                self.win_length, self.window)
         mag_y = stft_magnitude(y, self.fft_size, self.hop_size,
                                self.win_length, self.window)
+        if _ext.use_hip(mag_x) and _ext.ext() is not None and \
+                hasattr(_ext.ext(), "stft_loss_reduce"):
+            from melgan_amd.ops.fused import fused_stft_pair_loss
+
+            return fused_stft_pair_loss(mag_x, mag_y.detach())
         sc = torch.norm(mag_y - mag_x, p="fro") / \
             torch.norm(mag_y, p="fro").clamp_min(1e-7)
         log_mag = F.l1_loss(torch.log(mag_x), torch.log(mag_y))
