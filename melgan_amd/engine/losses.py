@@ -23,14 +23,28 @@ def adversarial_g_loss(disc_fake) -> torch.Tensor:
 
 
 def feature_matching_loss(disc_fake, disc_real) -> torch.Tensor:
-    """L1 over all intermediate feature maps (scores excluded)."""
-    loss = 0.0
-    n = 0
+    """L1 over all intermediate feature maps (scores excluded).
+    On GPU: one multi-tensor kernel each way (K11b) instead of ~60
+    elementwise/reduce launches."""
+    flat_f, flat_r = [], []
     for feats_f, feats_r in zip(disc_fake, disc_real):
         for f, r in zip(feats_f[:-1], feats_r[:-1]):
-            loss = loss + F.l1_loss(f.float(), r.float().detach())
-            n += 1
-    return loss / max(n, 1)
+            flat_f.append(f)
+            flat_r.append(r.detach())
+    if not flat_f:
+        return torch.zeros(())
+    from melgan_amd.ops import _ext
+
+    if _ext.use_hip(flat_f[0]) and _ext.ext() is not None and \
+            hasattr(_ext.ext(), "multi_l1_fwd") and \
+            all(t.dtype == flat_f[0].dtype for t in flat_f + flat_r):
+        from melgan_amd.ops.fused import fused_feature_matching
+
+        return fused_feature_matching(flat_f, flat_r)
+    loss = 0.0
+    for f, r in zip(flat_f, flat_r):
+        loss = loss + F.l1_loss(f.float(), r.float())
+    return loss / len(flat_f)
 
 
 def discriminator_loss(disc_real, disc_fake) -> torch.Tensor:

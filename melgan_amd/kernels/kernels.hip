@@ -481,6 +481,10 @@ torch::Tensor avg_pool1d_421_bwd(torch::Tensor gy, int64_t Tin);
 torch::Tensor stft_loss_reduce(torch::Tensor mx, torch::Tensor my);
 torch::Tensor stft_loss_bwd(torch::Tensor mx, torch::Tensor my,
                             torch::Tensor sums, torch::Tensor gouts);
+torch::Tensor multi_l1_fwd(torch::Tensor tbl, torch::Tensor chunks,
+                           int64_t nt, torch::ScalarType dt);
+void multi_l1_bwd(torch::Tensor tbl, torch::Tensor chunks, int64_t nt,
+                  torch::Tensor gout, torch::ScalarType dt);
 
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
@@ -872,6 +876,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avg_pool1d_421_bwd", &avg_pool1d_421_bwd);
   m.def("stft_loss_reduce", &stft_loss_reduce);
   m.def("stft_loss_bwd", &stft_loss_bwd);
+  m.def("multi_l1_fwd", &multi_l1_fwd);
+  m.def("multi_l1_bwd", &multi_l1_bwd);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

@@ -247,6 +247,61 @@ __global__ void stft_loss_bwd_kernel(const float* __restrict__ mx,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K11b: multi-tensor feature-matching L1 over ~20 (fake, real) feature
+// pairs: ONE forward reduction launch, ONE backward launch.
+// chunk_map packs (tensor_idx << 48) | elem_offset (adam-style table).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void multi_l1_fwd_kernel(T** __restrict__ f_ptrs,
+                                    T** __restrict__ r_ptrs,
+                                    const int64_t* __restrict__ numels,
+                                    const int64_t* __restrict__ chunk_map,
+                                    float* __restrict__ partial, int nt) {
+  __shared__ float red[256];
+  const int64_t packed = chunk_map[blockIdx.x];
+  const int ti = (int)(packed >> 48);
+  const int64_t off = packed & ((1LL << 48) - 1);
+  const int64_t n = numels[ti];
+  const T* f = f_ptrs[ti] + off;
+  const T* r = r_ptrs[ti] + off;
+  const int64_t len = min((int64_t)(1 << 16), n - off);
+  // per-tensor mean then mean over tensors: weight 1/(nt * n)
+  const float wgt = 1.f / ((float)nt * (float)n);
+  float acc = 0.f;
+  for (int64_t i = threadIdx.x; i < len; i += blockDim.x)
+    acc += fabsf(static_cast<float>(f[i]) - static_cast<float>(r[i]));
+  red[threadIdx.x] = acc * wgt;
+  __syncthreads();
+  for (int o = 128; o > 0; o >>= 1) {
+    if (threadIdx.x < o) red[threadIdx.x] += red[threadIdx.x + o];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(partial, red[0]);
+}
+
+template <typename T>
+__global__ void multi_l1_bwd_kernel(T** __restrict__ f_ptrs,
+                                    T** __restrict__ r_ptrs,
+                                    T** __restrict__ g_ptrs,
+                                    const int64_t* __restrict__ numels,
+                                    const int64_t* __restrict__ chunk_map,
+                                    const float* __restrict__ gout, int nt) {
+  const int64_t packed = chunk_map[blockIdx.x];
+  const int ti = (int)(packed >> 48);
+  const int64_t off = packed & ((1LL << 48) - 1);
+  const int64_t n = numels[ti];
+  const T* f = f_ptrs[ti] + off;
+  const T* r = r_ptrs[ti] + off;
+  T* g = g_ptrs[ti] + off;
+  const int64_t len = min((int64_t)(1 << 16), n - off);
+  const float scale = gout[0] / ((float)nt * (float)n);
+  for (int64_t i = threadIdx.x; i < len; i += blockDim.x) {
+    const float d = static_cast<float>(f[i]) - static_cast<float>(r[i]);
+    g[i] = static_cast<T>(d > 0.f ? scale : (d < 0.f ? -scale : 0.f));
+  }
+}
+
 }  // namespace melgan
 
 using namespace melgan;
@@ -359,7 +414,8 @@ torch::Tensor stft_loss_reduce(torch::Tensor mx, torch::Tensor my) {
   auto out = torch::zeros({3}, mx.options());
   const int64_t n = mx.numel();
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(stft_loss_reduce_kernel, dim3(gblocks(n)), dim3(256),
+  const int rb = (int)std::min<int64_t>(512, (n + 255) / 256);
+  hipLaunchKernelGGL(stft_loss_reduce_kernel, dim3(rb), dim3(256),
                      0, stream, mx.data_ptr<float>(), my.data_ptr<float>(),
                      out.data_ptr<float>(), n);
   return out;
@@ -375,4 +431,38 @@ torch::Tensor stft_loss_bwd(torch::Tensor mx, torch::Tensor my,
                      sums.data_ptr<float>(), gouts.data_ptr<float>(),
                      gx.data_ptr<float>(), n);
   return gx;
+}
+
+// table layout: [f*nt | r*nt | g*nt | numel*nt] int64 on device
+torch::Tensor multi_l1_fwd(torch::Tensor tbl, torch::Tensor chunks,
+                           int64_t nt, torch::ScalarType dt) {
+  auto out = torch::zeros({1}, torch::TensorOptions()
+                                   .dtype(torch::kFloat)
+                                   .device(tbl.device()));
+  int64_t* tp = tbl.data_ptr<int64_t>();
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, dt, "multi_l1_fwd", [&] {
+        hipLaunchKernelGGL((multi_l1_fwd_kernel<scalar_t>),
+                           dim3((int)chunks.numel()), dim3(256), 0, stream,
+                           (scalar_t**)tp, (scalar_t**)(tp + nt),
+                           tp + 3 * nt, chunks.data_ptr<int64_t>(),
+                           out.data_ptr<float>(), (int)nt);
+      });
+  return out;
+}
+
+void multi_l1_bwd(torch::Tensor tbl, torch::Tensor chunks, int64_t nt,
+                  torch::Tensor gout, torch::ScalarType dt) {
+  int64_t* tp = tbl.data_ptr<int64_t>();
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, dt, "multi_l1_bwd", [&] {
+        hipLaunchKernelGGL((multi_l1_bwd_kernel<scalar_t>),
+                           dim3((int)chunks.numel()), dim3(256), 0, stream,
+                           (scalar_t**)tp, (scalar_t**)(tp + nt),
+                           (scalar_t**)(tp + 2 * nt), tp + 3 * nt,
+                           chunks.data_ptr<int64_t>(),
+                           gout.data_ptr<float>(), (int)nt);
+      });
 }

@@ -327,3 +327,53 @@ class _AvgPool421Fn(torch.autograd.Function):
 
 def fused_avg_pool421(x):
     return _AvgPool421Fn.apply(x)
+
+
+def _l1_table(feats_f, feats_r, grads):
+    nt = len(feats_f)
+    CHUNK = 1 << 16
+    ptrs = [t.data_ptr() for t in feats_f] + [t.data_ptr() for t in feats_r]
+    ptrs += [g.data_ptr() for g in grads] if grads is not None \
+        else [0] * nt
+    host = ptrs + [t.numel() for t in feats_f]
+    chunks = []
+    for i, t in enumerate(feats_f):
+        for off in range(0, t.numel(), CHUNK):
+            chunks.append((i << 48) | off)
+    dev = feats_f[0].device
+    tbl = torch.tensor(host, dtype=torch.long).to(dev, non_blocking=True)
+    ch = torch.tensor(chunks, dtype=torch.long).to(dev, non_blocking=True)
+    return tbl, ch
+
+
+class _MultiL1Fn(torch.autograd.Function):
+    """Feature-matching L1 over ~20 (fake, real) feature-map pairs in
+    ONE kernel launch each way (K11b)."""
+
+    @staticmethod
+    def forward(ctx, n, *tensors):
+        C = _ext.require_ext()
+        feats_f = [t.contiguous() for t in tensors[:n]]
+        feats_r = [t.contiguous() for t in tensors[n:]]
+        ctx.save_for_backward(*feats_f, *feats_r)
+        ctx.n = n
+        tbl, ch = _l1_table(feats_f, feats_r, None)
+        out = C.multi_l1_fwd(tbl, ch, n, feats_f[0].dtype)
+        return out[0]
+
+    @staticmethod
+    def backward(ctx, gout):
+        C = _ext.require_ext()
+        n = ctx.n
+        ts = ctx.saved_tensors
+        feats_f = list(ts[:n])
+        feats_r = list(ts[n:])
+        grads = [torch.empty_like(f) for f in feats_f]
+        tbl, ch = _l1_table(feats_f, feats_r, grads)
+        C.multi_l1_bwd(tbl, ch, n, gout.detach().float().reshape(1)
+                       .contiguous(), feats_f[0].dtype)
+        return (None, *grads, *([None] * n))
+
+
+def fused_feature_matching(feats_f, feats_r):
+    return _MultiL1Fn.apply(len(feats_f), *feats_f, *feats_r)

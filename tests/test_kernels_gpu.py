@@ -625,3 +625,34 @@ def test_fused_stft_pair_loss_grad():
     assert abs(float(sc1) - float(sc2)) < 1e-4
     assert abs(float(mag1) - float(mag2)) < 1e-4
     _cmp(g1, x2.grad, rtol=1e-3, atol=1e-4, what="stft pair loss gx")
+
+
+def test_fused_feature_matching():
+    """K11b multi-tensor L1: value and grads vs composite."""
+    from melgan_amd.engine.losses import feature_matching_loss
+
+    torch.manual_seed(0)
+    fake = [[torch.randn(2, c, 64, device=DEV, requires_grad=True)
+             for c in (16, 64, 256)] + [torch.randn(2, 1, 8, device=DEV)]
+            for _ in range(2)]
+    real = [[torch.randn_like(f) for f in feats] for feats in fake]
+    loss = feature_matching_loss(fake, real)
+    loss.backward()
+    g_fused = [f.grad.clone() for feats in fake for f in feats[:-1]]
+    for feats in fake:
+        for f in feats[:-1]:
+            f.grad = None
+
+    ref = 0.0
+    n = 0
+    for feats_f, feats_r in zip(fake, real):
+        for f, r in zip(feats_f[:-1], feats_r[:-1]):
+            ref = ref + torch.nn.functional.l1_loss(f.float(),
+                                                    r.float().detach())
+            n += 1
+    ref = ref / n
+    ref.backward()
+    g_ref = [f.grad for feats in fake for f in feats[:-1]]
+    assert abs(float(loss) - float(ref)) < 1e-4
+    for a, b in zip(g_fused, g_ref):
+        _cmp(a, b, rtol=1e-3, atol=1e-5, what="fm grads")
