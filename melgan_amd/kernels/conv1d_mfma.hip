@@ -441,13 +441,15 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
-// phase-decomposed strided backward-data (zero-pad geometry only):
-// outputs j with (j+p) % s == phi form one GEMM per phase:
-//   gx[b][ci][j=u*s+phi-p] = sum_{co in grp, i} w[co][cig][phi+i*s] *
-//                            gz[b][co][u-i]
-// No zero-stuffing (s x less MFMA work than the stuffed path) and no
-// host-side weight flip.  16-row M tile (grouped convs have Cpg<=16).
-// grid: (s * ceil(Umax/128), G * ceil(Cpg/16), B)
+// phase-decomposed strided backward-data (zero-pad geometry only).
+// KEY IDENTITY: for output j = u*s + phi - p the contributing gz index
+// t = (j + p - k)/s with k = phi + i*s is  t = u - i  — INDEPENDENT of
+// the phase phi.  So the M rows of one MFMA tile are (cig, phi) pairs
+// sharing one B operand: for the k=41 discriminator towers
+// (Cpg=4, s=4) the 16-row tile is EXACTLY filled — zero masked waste.
+//   gx[b][g*Cpg+cig][u*s+phi-p] = sum_{co,i} w[g*Copg+co][cig][phi+i*s]
+//                                 * gz[b][g*Copg+co][u-i]
+// grid: (ceil(Umax/128), G * ceil(Cpg*s/16), B)
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(256)
 convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
@@ -457,21 +459,20 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
                         int pad, int groups, int utiles) {
   __shared__ __attribute__((aligned(16))) unsigned short smem[
       16 * LDW + 128 * LDW];
-  unsigned short* lds_w = smem;              // [16 cig][kk]
+  unsigned short* lds_w = smem;              // [16 (cig,phi)][kk]
   unsigned short* lds_gs = smem + 16 * LDW;  // [col=u(128)][kk]
 
   const int Copg = Cout / groups;
   const int Cpg = Cin / groups;
-  const int mtiles = (Cpg + 15) / 16;
+  const int rows_pg = Cpg * stride;          // (cig, phi) rows per group
+  const int mtiles = (rows_pg + 15) / 16;
   const int g = blockIdx.y / mtiles;
   const int mt = blockIdx.y - g * mtiles;
-  const int ci_base = g * Cpg + mt * 16;
-  const int ci_end = (g + 1) * Cpg;
-  const int phi = blockIdx.x / utiles;       // output phase
-  const int u0 = (blockIdx.x - phi * utiles) * 128;
+  const int row_base = mt * 16;
+  const int u0 = blockIdx.x * 128;
   const int b = blockIdx.z;
-  const int nk = (K - phi + stride - 1) / stride;  // taps for this phase
-  const int Npg = Copg * nk;
+  const int nkmax = (K + stride - 1) / stride;
+  const int Npg = Copg * nkmax;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -483,18 +484,21 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
 
   for (int nn0 = 0; nn0 < Npg; nn0 += CI_CH) {
     for (int idx = tid; idx < 16 * CI_CH; idx += 256) {
-      const int cig = idx / CI_CH;
-      const int kk = idx - cig * CI_CH;
+      const int r = idx / CI_CH;
+      const int kk = idx - r * CI_CH;
       const int nn = nn0 + kk;
+      const int rg = row_base + r;           // (cig, phi): phi fastest
       unsigned short v = 0;
-      if (ci_base + cig < ci_end && nn < Npg) {
-        const int co = nn / nk;
-        const int i = nn - co * nk;
+      if (nn < Npg && rg < rows_pg) {
+        const int cig = rg / stride;
+        const int phi = rg - cig * stride;
+        const int co = nn / nkmax;
+        const int i = nn - co * nkmax;
         const int k = phi + i * stride;
-        v = w[((int64_t)(g * Copg + co) * Cpg + (ci_base + cig) -
-               g * Cpg) * K + k];
+        if (k < K)
+          v = w[((int64_t)(g * Copg + co) * Cpg + cig) * K + k];
       }
-      lds_w[cig * LDW + kk] = v;
+      lds_w[r * LDW + kk] = v;
     }
     for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
       const int col = idx / CI_CH;
@@ -502,8 +506,8 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
       const int nn = nn0 + kk;
       unsigned short v = 0;
       if (nn < Npg) {
-        const int co = nn / nk;
-        const int i = nn - co * nk;
+        const int co = nn / nkmax;
+        const int i = nn - co * nkmax;
         const int t = u0 + col - i;
         if (t >= 0 && t < To)
           v = gz[((int64_t)b * Cout + g * Copg + co) * To + t];
@@ -526,191 +530,18 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
   const int row4 = (lane >> 4) * 4;
   for (int n = 0; n < 2; ++n) {
     const int u = u0 + wid * 32 + n * 16 + col;
-    const int j = u * stride + phi - pad;
-    if (j < 0 || j >= Tin) continue;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int ci = ci_base + row4 + r;
-      if (ci >= ci_end || ci >= Cin) continue;
+      const int rg = row_base + row4 + r;
+      if (rg >= rows_pg) continue;
+      const int cig = rg / stride;
+      const int phi = rg - cig * stride;
+      const int j = u * stride + phi - pad;
+      if (j < 0 || j >= Tin) continue;
+      const int ci = g * Cpg + cig;
       union { __bf16 bb; unsigned short s; } cv{(__bf16)acc[n][r]};
       gx[((int64_t)b * Cin + ci) * Tin + j] = cv.s;
     }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// deep-contraction / small-T forward (disc k=5 etc.):
-// batch folded into the column dimension (N = B*To), 256-column tiles,
-// fused (ci,k) contraction — 4x more output per staging round than the
-// 64-column kernels, which are staging/barrier-bound at these shapes.
-// stride 1, groups 1.  grid: (ceil(B*To/256), ceil(Cout/64), 1)
-// ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(256)
-conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
-                            const unsigned short* __restrict__ w,
-                            const float* __restrict__ bias,
-                            unsigned short* __restrict__ y,
-                            float* __restrict__ yws, int B, int Cin,
-                            int Tin, int Cout, int To, int K, int dil,
-                            int pad, bool reflect, int act, float slope,
-                            bool act_before, int l2cin, int l2to) {
-  __shared__ __attribute__((aligned(16))) unsigned short smem[
-      (64 + 256) * LDW];
-  unsigned short* lds_w = smem;               // [64co][kk]
-  unsigned short* lds_xs = smem + 64 * LDW;   // [col][kk]
-
-  const int cg0 = blockIdx.x * 256;           // global (b,t) column origin
-  const int co0 = blockIdx.y * BM;
-  const int NK = Cin * K;
-  const int64_t NCOL = (int64_t)B * To;
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int wr = wid >> 1, wc = wid & 1;      // wc: 128-col half
-  const int row = lane & 15;
-  const int kk0 = (lane >> 4) * 8;
-
-  f32x4 acc[2][8] = {};
-
-  // split-K over gridDim.z: this block covers chunk indices
-  // [z*per, min((z+1)*per, nchunks))
-  const int nchunks = (NK + CI_CH - 1) / CI_CH;
-  const int per = (nchunks + gridDim.z - 1) / gridDim.z;
-  const int c_lo = blockIdx.z * per;
-  const int c_hi = min(c_lo + per, nchunks);
-
-  for (int cix = c_lo; cix < c_hi; ++cix) {
-    const int nn0 = cix * CI_CH;
-    for (int idx = tid; idx < BM * CI_CH; idx += 256) {
-      const int co = idx / CI_CH;
-      const int kk = idx - co * CI_CH;
-      const int nn = nn0 + kk;
-      unsigned short v = 0;
-      if (co0 + co < Cout && nn < NK) {
-        int ci, k;
-        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
-        else { ci = nn % Cin; k = nn / Cin; }
-        v = w[((int64_t)(co0 + co) * Cin + ci) * K + k];
-      }
-      lds_w[co * LDW + kk] = v;
-    }
-    for (int idx = tid; idx < 256 * CI_CH; idx += 256) {
-      const int col = idx / CI_CH;
-      const int kk = idx - col * CI_CH;
-      const int nn = nn0 + kk;
-      unsigned short v = 0;
-      const int64_t cg = cg0 + col;
-      if (nn < NK && cg < NCOL) {
-        int b, t;
-        if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
-        else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
-        int ci, k;
-        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
-        else { ci = nn % Cin; k = nn / Cin; }
-        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-        v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
-        if (act_before) v = leaky_bf16(v, slope);
-      }
-      lds_xs[col * LDW + kk] = v;
-    }
-    __syncthreads();
-
-    {
-      bf16x8 a0 = *(const bf16x8*)(lds_w + (wr * 32 + row) * LDW + kk0);
-      bf16x8 a1 = *(const bf16x8*)(lds_w + (wr * 32 + 16 + row) * LDW + kk0);
-#pragma unroll
-      for (int n = 0; n < 8; ++n) {
-        bf16x8 bn = *(const bf16x8*)(lds_xs +
-                                     (wc * 128 + n * 16 + row) * LDW + kk0);
-        acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bn, acc[0][n], 0, 0, 0);
-        acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bn, acc[1][n], 0, 0, 0);
-      }
-    }
-    __syncthreads();
-  }
-
-  const int col = lane & 15;
-  const int row4 = (lane >> 4) * 4;
-  for (int m = 0; m < 2; ++m)
-    for (int n = 0; n < 8; ++n) {
-      const int64_t cg = cg0 + wc * 128 + n * 16 + col;
-      if (cg >= NCOL) continue;
-      int b, t;
-      if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
-      else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int co = co0 + wr * 32 + m * 16 + row4 + r;
-        if (co >= Cout) continue;
-        const int64_t oi = ((int64_t)b * Cout + co) * To + t;
-        if (gridDim.z == 1) {
-          float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
-          if (!act_before) o = apply_act(o, act, slope);
-          union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-          y[oi] = cv.s;
-        } else {
-          atomicAdd(&yws[oi], acc[m][n][r]);
-        }
-      }
-    }
-}
-
-// finalize after split-K: y = act(ws + bias)
-__global__ void deep_finalize_kernel(const float* __restrict__ yws,
-                                     const float* __restrict__ bias,
-                                     unsigned short* __restrict__ y, int B,
-                                     int Cout, int To, int act, float slope,
-                                     bool act_before) {
-  const int64_t nall = (int64_t)B * Cout * To;
-  const int64_t step = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nall;
-       i += step) {
-    const int co = (int)((i / To) % Cout);
-    float o = yws[i] + (bias ? bias[co] : 0.f);
-    if (!act_before) o = apply_act(o, act, slope);
-    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-    y[i] = cv.s;
-  }
-}
-
-// dot-form forward for tiny Cout (disc score conv 1024->1 k3): one block
-// per (b, t-tile); 256 threads reduce over Cin*K via LDS tree.
-__global__ void __launch_bounds__(256)
-conv1d_fwd_dot_kernel(const unsigned short* __restrict__ x,
-                      const unsigned short* __restrict__ w,
-                      const float* __restrict__ bias,
-                      unsigned short* __restrict__ y, int B, int Cin,
-                      int Tin, int Cout, int To, int K, int dil, int pad,
-                      bool reflect, int act, float slope, bool act_before) {
-  __shared__ float red[256];
-  const int t = blockIdx.x;
-  const int co = blockIdx.y;
-  const int b = blockIdx.z;
-  const int NK = Cin * K;
-  float acc = 0.f;
-  for (int nn = threadIdx.x; nn < NK; nn += 256) {
-    const int ci = nn / K;
-    const int k = nn - ci * K;
-    const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-    unsigned short v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
-    if (act_before) v = leaky_bf16(v, slope);
-    union { unsigned short s; __bf16 bb; } xv{v};
-    union { unsigned short s; __bf16 bb; } wv{
-        w[((int64_t)co * Cin + ci) * K + k]};
-    acc = fmaf((float)wv.bb, (float)xv.bb, acc);
-  }
-  red[threadIdx.x] = acc;
-  __syncthreads();
-  for (int off = 128; off > 0; off >>= 1) {
-    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    float o = red[0] + (bias ? bias[co] : 0.f);
-    if (!act_before) o = apply_act(o, act, slope);
-    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-    y[((int64_t)b * Cout + co) * To + t] = cv.s;
   }
 }
 
@@ -1177,13 +1008,14 @@ torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
   const int pad2 = (K - 1) * (int)dil;
   const int Tpad = (int)Tin + 2 * (int)pad;
   const int CpgX = Cpg;  // (per-group input channels == gx channels/group)
-  if (!reflect && stride > 1 && dil == 1 && CpgX <= 16) {
+  if (!reflect && stride > 1 && dil == 1 &&
+      CpgX * stride <= 64) {  // (cig,phi) rows per group fit few m-tiles
     auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
     // Umax covers all j in [0, Tin): u <= (Tin - 1 + pad)/stride
     const int Umax = ((int)Tin - 1 + (int)pad) / (int)stride + 1;
     const int utiles = cdiv_i(Umax, 128);
-    dim3 grid((int)stride * utiles,
-              (int)groups * cdiv_i(CpgX, 16), B);
+    dim3 grid(utiles,
+              (int)groups * cdiv_i(CpgX * (int)stride, 16), B);
     auto stream2 = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(convbwd_phased16_kernel, grid, dim3(256), 0, stream2,
                        (const unsigned short*)gz.data_ptr(),
