@@ -501,8 +501,8 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
       lds_w[r * LDW + kk] = v;
     }
     for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
-      const int col = idx / CI_CH;
-      const int kk = idx - col * CI_CH;
+      const int kk = idx / 128;        // consecutive tid -> consecutive u
+      const int col = idx - kk * 128;  // (coalesced gz reads per row)
       const int nn = nn0 + kk;
       unsigned short v = 0;
       if (nn < Npg) {
