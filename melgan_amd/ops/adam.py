@@ -57,6 +57,12 @@ class FusedAdam(torch.optim.Optimizer):
                 e.multi_tensor_adam_tbl(
                     tbl, chunks, nt, group["lr"], beta1, beta2,
                     group["eps"], group["weight_decay"], steps[0])
+                # the kernel writes params through raw pointers, which
+                # does NOT bump torch's _version counter — bump an
+                # explicit epoch so weight-norm fold caches (keyed on it)
+                # invalidate (frozen-weights bug otherwise!)
+                for p in params:
+                    p._wn_epoch = getattr(p, "_wn_epoch", 0) + 1
             else:
                 # torch's own multi-tensor path (same math)
                 torch._foreach_mul_(exp_avgs, beta1)

@@ -38,8 +38,15 @@ def _wn_fold(v: torch.Tensor, g: torch.Tensor, dtype: torch.dtype):
 
 
 def _wn_cached(mod, dtype: torch.dtype):
-    """Fold w=g*v/||v|| once per optimizer step (param _version keyed)."""
-    key = (mod.weight_v._version, mod.weight_g._version, dtype)
+    """Fold w=g*v/||v|| once per optimizer step.
+
+    Cache key: torch's _version counter (bumped by any in-place torch
+    op, e.g. load_state_dict or eager optimizers) PLUS an explicit
+    _wn_epoch attribute bumped by FusedAdam (whose kernel writes params
+    through raw pointers and therefore does not touch _version)."""
+    v, g = mod.weight_v, mod.weight_g
+    key = (v._version, g._version, getattr(v, "_wn_epoch", 0),
+           getattr(g, "_wn_epoch", 0), dtype)
     cache = getattr(mod, "_wn_cache", None)
     if cache is not None and cache[0] == key:
         return cache[1], cache[2]

@@ -656,3 +656,30 @@ def test_fused_feature_matching():
     assert abs(float(loss) - float(ref)) < 1e-4
     for a, b in zip(g_fused, g_ref):
         _cmp(a, b, rtol=1e-3, atol=1e-5, what="fm grads")
+
+
+def test_weights_actually_train_with_fused_adam():
+    """Regression: the wn-fold cache must invalidate after FusedAdam's
+    in-place kernel update (raw pointer writes do not bump _version);
+    with a stale cache the fused forward freezes and only biases learn."""
+    from melgan_amd.ops.adam import FusedAdam
+
+    mod = _mkconv(32, 32, 3, pad_mode="reflect", act=None)
+    opt = FusedAdam(mod.parameters(), lr=1e-2, betas=(0.5, 0.9))
+    from melgan_amd.ops.fused import fused_conv1d
+
+    x = torch.randn(2, 32, 128, device=DEV)
+    y0 = fused_conv1d(mod, x)
+    for _ in range(3):
+        opt.zero_grad(set_to_none=True)
+        y = fused_conv1d(mod, x)
+        y.pow(2).mean().backward()
+        opt.step()
+    with torch.no_grad():
+        y1 = fused_conv1d(mod, x)
+    # remove the bias effect: bias changes shift y uniformly; compare
+    # de-meaned outputs, which only change if the WEIGHTS moved
+    d0 = y0 - y0.mean(dim=2, keepdim=True)
+    d1 = y1 - y1.mean(dim=2, keepdim=True)
+    rel = (d1 - d0).abs().mean() / d0.abs().mean()
+    assert rel > 1e-3, f"weights frozen (de-meaned output moved {rel:.2e})"
