@@ -4,22 +4,25 @@ Kernel API contract (implemented in melgan_amd/kernels/):
 
   conv1d_fwd(x, w, bias?, stride, dilation, pad, reflect, groups,
              act, slope, act_before) -> y
-      direct conv; reflection/zero pad folded into addressing; LeakyReLU
-      prologue (act_before) or LeakyReLU/Tanh epilogue fused.
-  conv1d_bwd_data(gz, w, stride, dilation, pad, reflect, groups, T_in) -> gx
-  conv1d_bwd_weight(gz, x, stride, dilation, pad, reflect, groups, K,
-                    act_before, slope) -> gw (fp32)
-  conv1d_bias_grad(gz) -> gbias (fp32)
-  gather_conv_fwd(x, w, bias?, stride_up, pad, act_slope, act_before,
-                  tanh_out) -> y      # transposed conv, output-stationary
-  gather_conv_bwd_data / gather_conv_bwd_weight
-  act_bwd(gy, y, act, slope) -> gz    # epilogue activation backward
-  leaky_bwd_inplace(gx, x, slope)     # prologue activation backward
-  multi_tensor_adam(...)
+      direct conv; zero/reflection pad folded into addressing; LeakyReLU
+      prologue (act_before) or LeakyReLU/Tanh epilogue fused; dispatches
+      internally across the MFMA tiers (specialized stride-1, generic
+      grouped, 16-row small-group, dot-form tiny-Cout) and the fp32 VALU
+      tier.
+  conv1d_bwd_data / conv1d_bwd_weight      (MFMA GEMM forms + VALU)
+  conv_transpose1d_fwd / _bwd_data / _bwd_weight
+      (zero-stuff + weight-flip reductions onto the same conv engine)
+  wn_fold / wn_bwd                         (weight-norm fold, 1 kernel)
+  act_bwd / leaky_bwd_inplace / conv1d_bias_grad
+  stft_frame / stft_frame_bwd / complex_mag_clamp / mag_bwd
+  stft_loss_reduce / stft_loss_bwd         (fused SC+log-mag loss)
+  multi_l1_fwd / multi_l1_bwd              (feature-matching L1)
+  avg_pool1d_421 (+_bwd), multi_tensor_adam_tbl
 
-Weight-norm chain rule (w = g*v/||v||) is applied HERE in fp32 (the
-per-channel reductions are tiny next to the convs — SURVEY.md §2.2 K5
-is eliminated from the hot path by folding w into the conv kernels).
+Weight-norm chain rule (w = g*v/||v||) is applied HERE in fp32; the
+fold is cached per optimizer step (see _wn_cached — the cache key
+includes the explicit _wn_epoch that FusedAdam bumps, because its
+kernel's raw-pointer writes do not touch torch's _version counter).
 """
 
 from __future__ import annotations
