@@ -545,6 +545,182 @@ convbwd_phased16_kernel(const unsigned short* __restrict__ gz,
   }
 }
 
+// ---------------------------------------------------------------------------
+// deep-contraction / small-T forward (disc k=5 etc.):
+// batch folded into the column dimension (N = B*To), 256-column tiles,
+// fused (ci,k) contraction — 4x more output per staging round than the
+// 64-column kernels, which are staging/barrier-bound at these shapes.
+// stride 1, groups 1.  grid: (ceil(B*To/256), ceil(Cout/64), 1)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+conv1d_fwd_mfma_deep_kernel(const unsigned short* __restrict__ x,
+                            const unsigned short* __restrict__ w,
+                            const float* __restrict__ bias,
+                            unsigned short* __restrict__ y,
+                            float* __restrict__ yws, int B, int Cin,
+                            int Tin, int Cout, int To, int K, int dil,
+                            int pad, bool reflect, int act, float slope,
+                            bool act_before, int l2cin, int l2to) {
+  __shared__ __attribute__((aligned(16))) unsigned short smem[
+      (64 + 256) * LDW];
+  unsigned short* lds_w = smem;               // [64co][kk]
+  unsigned short* lds_xs = smem + 64 * LDW;   // [col][kk]
+
+  const int cg0 = blockIdx.x * 256;           // global (b,t) column origin
+  const int co0 = blockIdx.y * BM;
+  const int NK = Cin * K;
+  const int64_t NCOL = (int64_t)B * To;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1, wc = wid & 1;      // wc: 128-col half
+  const int row = lane & 15;
+  const int kk0 = (lane >> 4) * 8;
+
+  f32x4 acc[2][8] = {};
+
+  // split-K over gridDim.z: this block covers chunk indices
+  // [z*per, min((z+1)*per, nchunks))
+  const int nchunks = (NK + CI_CH - 1) / CI_CH;
+  const int per = (nchunks + gridDim.z - 1) / gridDim.z;
+  const int c_lo = blockIdx.z * per;
+  const int c_hi = min(c_lo + per, nchunks);
+
+  for (int cix = c_lo; cix < c_hi; ++cix) {
+    const int nn0 = cix * CI_CH;
+    for (int idx = tid; idx < BM * CI_CH; idx += 256) {
+      const int co = idx / CI_CH;
+      const int kk = idx - co * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      if (co0 + co < Cout && nn < NK) {
+        int ci, k;
+        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
+        else { ci = nn % Cin; k = nn / Cin; }
+        v = w[((int64_t)(co0 + co) * Cin + ci) * K + k];
+      }
+      lds_w[co * LDW + kk] = v;
+    }
+    for (int idx = tid; idx < 256 * CI_CH; idx += 256) {
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
+      const int nn = nn0 + kk;
+      unsigned short v = 0;
+      const int64_t cg = cg0 + col;
+      if (nn < NK && cg < NCOL) {
+        int b, t;
+        if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
+        else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
+        int ci, k;
+        if (l2cin >= 0) { ci = nn & (Cin - 1); k = nn >> l2cin; }
+        else { ci = nn % Cin; k = nn / Cin; }
+        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+        v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
+        if (act_before) v = leaky_bf16(v, slope);
+      }
+      lds_xs[col * LDW + kk] = v;
+    }
+    __syncthreads();
+
+    {
+      bf16x8 a0 = *(const bf16x8*)(lds_w + (wr * 32 + row) * LDW + kk0);
+      bf16x8 a1 = *(const bf16x8*)(lds_w + (wr * 32 + 16 + row) * LDW + kk0);
+#pragma unroll
+      for (int n = 0; n < 8; ++n) {
+        bf16x8 bn = *(const bf16x8*)(lds_xs +
+                                     (wc * 128 + n * 16 + row) * LDW + kk0);
+        acc[0][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bn, acc[0][n], 0, 0, 0);
+        acc[1][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bn, acc[1][n], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+  for (int m = 0; m < 2; ++m)
+    for (int n = 0; n < 8; ++n) {
+      const int64_t cg = cg0 + wc * 128 + n * 16 + col;
+      if (cg >= NCOL) continue;
+      int b, t;
+      if (l2to >= 0) { b = (int)(cg >> l2to); t = (int)cg & (To - 1); }
+      else { b = (int)(cg / To); t = (int)(cg - (int64_t)b * To); }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co0 + wr * 32 + m * 16 + row4 + r;
+        if (co >= Cout) continue;
+        const int64_t oi = ((int64_t)b * Cout + co) * To + t;
+        if (gridDim.z == 1) {
+          float o = acc[m][n][r] + (bias ? bias[co] : 0.f);
+          if (!act_before) o = apply_act(o, act, slope);
+          union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+          y[oi] = cv.s;
+        } else {
+          atomicAdd(&yws[oi], acc[m][n][r]);
+        }
+      }
+    }
+}
+
+// finalize after split-K: y = act(ws + bias)
+__global__ void deep_finalize_kernel(const float* __restrict__ yws,
+                                     const float* __restrict__ bias,
+                                     unsigned short* __restrict__ y, int B,
+                                     int Cout, int To, int act, float slope,
+                                     bool act_before) {
+  const int64_t nall = (int64_t)B * Cout * To;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nall;
+       i += step) {
+    const int co = (int)((i / To) % Cout);
+    float o = yws[i] + (bias ? bias[co] : 0.f);
+    if (!act_before) o = apply_act(o, act, slope);
+    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+    y[i] = cv.s;
+  }
+}
+
+// dot-form forward for tiny Cout (disc score conv 1024->1 k3): one block
+// per (b, t-tile); 256 threads reduce over Cin*K via LDS tree.
+__global__ void __launch_bounds__(256)
+conv1d_fwd_dot_kernel(const unsigned short* __restrict__ x,
+                      const unsigned short* __restrict__ w,
+                      const float* __restrict__ bias,
+                      unsigned short* __restrict__ y, int B, int Cin,
+                      int Tin, int Cout, int To, int K, int dil, int pad,
+                      bool reflect, int act, float slope, bool act_before) {
+  __shared__ float red[256];
+  const int t = blockIdx.x;
+  const int co = blockIdx.y;
+  const int b = blockIdx.z;
+  const int NK = Cin * K;
+  float acc = 0.f;
+  for (int nn = threadIdx.x; nn < NK; nn += 256) {
+    const int ci = nn / K;
+    const int k = nn - ci * K;
+    const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+    unsigned short v = load_pad_bf16(xr, t + k * dil - pad, Tin, reflect);
+    if (act_before) v = leaky_bf16(v, slope);
+    union { unsigned short s; __bf16 bb; } xv{v};
+    union { unsigned short s; __bf16 bb; } wv{
+        w[((int64_t)co * Cin + ci) * K + k]};
+    acc = fmaf((float)wv.bb, (float)xv.bb, acc);
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    float o = red[0] + (bias ? bias[co] : 0.f);
+    if (!act_before) o = apply_act(o, act, slope);
+    union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+    y[((int64_t)b * Cout + co) * To + t] = cv.s;
+  }
+}
+
 // zero-stuff along t: y[b][c][t*s] = x[b][c][t] (rest 0)
 __global__ void zero_stuff_kernel(const unsigned short* __restrict__ x,
                                   unsigned short* __restrict__ y,
