@@ -2,11 +2,11 @@
 set -x
 cd /root/repo
 mkdir -p gpurun_out
-timeout 600 python -m pytest tests/test_kernels_gpu.py -q -m gpu --timeout 300 > gpurun_out/gpu_tests20.log 2>&1
-echo "PYTEST: $?"; tail -2 gpurun_out/gpu_tests20.log
-timeout 400 python bench.py --steps 15 --warmup 5 --inference-rtf > gpurun_out/bench_fixed.log 2>&1
-echo "BENCH: $?"; tail -1 gpurun_out/bench_fixed.log | head -c 250; echo
-timeout 900 python - > gpurun_out/train1000_fixed.log 2>&1 <<'PYEOF'
+timeout 600 python -m pytest tests/test_kernels_gpu.py -q -m gpu --timeout 300 > gpurun_out/gpu_tests21.log 2>&1
+echo "PYTEST: $?"; tail -2 gpurun_out/gpu_tests21.log
+timeout 400 python bench.py --steps 15 --warmup 5 --inference-rtf > gpurun_out/bench_phasepack2.log 2>&1
+echo "BENCH: $?"; tail -1 gpurun_out/bench_phasepack2.log | head -c 250; echo
+timeout 900 python - > gpurun_out/train1000_pp.log 2>&1 <<'PYEOF'
 import torch, json
 from melgan_amd.config import HParam, merge_default
 from melgan_amd.engine.trainer import Trainer
@@ -47,8 +47,8 @@ print(json.dumps(hist))
 print("phases(ms):", {k: round(v,2) for k,v in tr2.last_phases.items()})
 print("stft 0:", hist[0][1], "-> 450:", hist[9][1], "-> 950:", hist[-1][1])
 PYEOF
-echo "T1K: $?"; tail -3 gpurun_out/train1000_fixed.log
-timeout 500 python - > gpurun_out/bf16_vs_fp32_fixed.log 2>&1 <<'PYEOF'
+echo "T1K: $?"; tail -3 gpurun_out/train1000_pp.log
+timeout 500 python - > gpurun_out/bf16_vs_fp32_pp.log 2>&1 <<'PYEOF'
 import torch, json, numpy as np
 from melgan_amd.config import HParam, merge_default
 from melgan_amd.engine.trainer import Trainer
@@ -79,4 +79,4 @@ print(json.dumps(curves))
 print("max rel dev:", float(np.abs(a-b).max()/np.abs(a).max()),
       "decreasing:", a[-1] < a[0], b[-1] < b[0])
 PYEOF
-echo "NUMERICS: $?"; tail -2 gpurun_out/bf16_vs_fp32_fixed.log
+echo "NUMERICS: $?"; tail -2 gpurun_out/bf16_vs_fp32_pp.log
