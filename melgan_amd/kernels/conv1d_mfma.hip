@@ -300,8 +300,8 @@ conv1d_fwd_mfma_g_kernel(const unsigned short* __restrict__ x,
       lds_w[co * LDW + kk] = v;
     }
     for (int idx = tid; idx < BN * CI_CH; idx += 256) {
-      const int kk = idx / BN;         // consecutive tid -> consecutive t
-      const int col = idx - kk * BN;   // (coalesced x reads per row)
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
       const int nn = nn0 + kk;
       unsigned short v = 0;
       if (nn < Npg && t0 + col < To) {
@@ -396,8 +396,8 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
       lds_w[co * LDW + kk] = v;
     }
     for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
-      const int kk = idx / 128;        // consecutive tid -> consecutive t
-      const int col = idx - kk * 128;  // (coalesced x reads per row)
+      const int col = idx / CI_CH;
+      const int kk = idx - col * CI_CH;
       const int nn = nn0 + kk;
       unsigned short v = 0;
       if (nn < Npg && t0 + col < To) {
