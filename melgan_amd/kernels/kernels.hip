@@ -389,7 +389,7 @@ __global__ void wn_bwd_kernel(const float* __restrict__ gw,
 // multi-tensor fused Adam (K12): one launch for ALL tensors.
 // chunk table: chunk_tensor[i] = tensor index, chunk_off[i] = elem offset.
 // ---------------------------------------------------------------------------
-constexpr int ADAM_CHUNK = 1 << 16;
+constexpr int ADAM_CHUNK = 1 << 13;  // ~900 blocks for a 7M-param model
 
 __global__ void multi_tensor_adam_kernel(
     float** __restrict__ p_ptrs, float** __restrict__ g_ptrs,

@@ -86,7 +86,7 @@ class FusedAdam(torch.optim.Optimizer):
         once pointers are static."""
         import torch as _t
 
-        CHUNK = 1 << 16
+        CHUNK = 1 << 13  # must match ADAM_CHUNK in kernels.hip
         ptrs = []
         for lst in (params, grads, exp_avgs, exp_avg_sqs):
             ptrs.extend(t.data_ptr() for t in lst)
