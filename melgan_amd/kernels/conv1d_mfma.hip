@@ -362,7 +362,7 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
                            int dil, int pad, bool reflect, int groups,
                            int act, float slope, bool act_before) {
   __shared__ __attribute__((aligned(16))) unsigned short smem[
-      16 * LDW + 256 * LDW];
+      16 * LDW + 128 * LDW];
   unsigned short* lds_w = smem;               // [16co][kk]
   unsigned short* lds_xs = smem + 16 * LDW;   // [col=t(128)][kk]
 
@@ -373,7 +373,7 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
   const int mt = blockIdx.y - g * mtiles;
   const int co_base = g * Copg + mt * 16;
   const int co_end = (g + 1) * Copg;
-  const int t0 = blockIdx.x * 256;
+  const int t0 = blockIdx.x * 128;
   const int b = blockIdx.z;
   const int Npg = Cpg * K;
 
@@ -383,7 +383,7 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
   const int row = lane & 15;
   const int kk0 = (lane >> 4) * 8;
 
-  f32x4 acc[4] = {};
+  f32x4 acc[2] = {};
 
   for (int nn0 = 0; nn0 < Npg; nn0 += CI_CH) {
     for (int idx = tid; idx < 16 * CI_CH; idx += 256) {
@@ -395,7 +395,7 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
         v = w[((int64_t)(co_base + co) * Cpg + nn / K) * K + nn % K];
       lds_w[co * LDW + kk] = v;
     }
-    for (int idx = tid; idx < 256 * CI_CH; idx += 256) {
+    for (int idx = tid; idx < 128 * CI_CH; idx += 256) {
       const int col = idx / CI_CH;
       const int kk = idx - col * CI_CH;
       const int nn = nn0 + kk;
@@ -414,20 +414,19 @@ conv1d_fwd_mfma_g16_kernel(const unsigned short* __restrict__ x,
     __syncthreads();
     {
       bf16x8 a0 = *(const bf16x8*)(lds_w + row * LDW + kk0);
-#pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        bf16x8 bn = *(const bf16x8*)(lds_xs +
-                                     (wid * 64 + n * 16 + row) * LDW + kk0);
-        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bn, acc[n], 0, 0, 0);
-      }
+      bf16x8 b0 = *(const bf16x8*)(lds_xs + (wid * 32 + row) * LDW + kk0);
+      bf16x8 b1 =
+          *(const bf16x8*)(lds_xs + (wid * 32 + 16 + row) * LDW + kk0);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[1], 0, 0, 0);
     }
     __syncthreads();
   }
 
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
-  for (int n = 0; n < 4; ++n) {
-    const int t = t0 + wid * 64 + n * 16 + col;
+  for (int n = 0; n < 2; ++n) {
+    const int t = t0 + wid * 32 + n * 16 + col;
     if (t >= To) continue;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -1106,7 +1105,7 @@ torch::Tensor conv1d_fwd_mfma_generic(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty({B, Cout, (int)To}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
   if (Copg <= 16) {
-    dim3 grid(cdiv_i(To, 256), (int)groups * cdiv_i(Copg, 16), B);
+    dim3 grid(cdiv_i(To, 128), (int)groups * cdiv_i(Copg, 16), B);
     hipLaunchKernelGGL(conv1d_fwd_mfma_g16_kernel, grid, dim3(256), 0,
                        stream, (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)w.data_ptr(), bias_p,
