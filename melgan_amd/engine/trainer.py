@@ -156,11 +156,24 @@ class Trainer:
         g_loss = stft_loss
         adv = feat = None
         if self.step >= hp.train.discriminator_train_start_steps:
-            with torch.autocast("cuda", dtype=self.amp_dtype,
-                                enabled=self.use_amp):
-                d_fake = self.discriminator(fake)
-                with torch.no_grad():
-                    d_real = self.discriminator(audio.float())
+            # Freeze discriminator PARAMS for the generator-phase
+            # adversarial forward: the g loss needs gradients THROUGH the
+            # discriminator to `fake`, not INTO its weights.  Without
+            # this, g_loss.backward() (a) wastes the disc weight-grad
+            # kernels and (b) fires the disc's DP bucket hooks with
+            # g-phase gradients, which would make the d-step's
+            # all-reduce ship STALE gradients at world_size>1.
+            for p in self.discriminator.parameters():
+                p.requires_grad_(False)
+            try:
+                with torch.autocast("cuda", dtype=self.amp_dtype,
+                                    enabled=self.use_amp):
+                    d_fake = self.discriminator(fake)
+                    with torch.no_grad():
+                        d_real = self.discriminator(audio.float())
+            finally:
+                for p in self.discriminator.parameters():
+                    p.requires_grad_(True)
             adv = L.adversarial_g_loss(d_fake)
             feat = L.feature_matching_loss(d_fake, d_real)
             g_loss = g_loss + self.lambda_adv * adv + self.lambda_feat * feat

@@ -138,6 +138,13 @@ class BucketAllReduceDP:
         if not self.enabled:
             return
         for b in self.buckets:
+            if b.ready not in (0, len(b.params)):
+                raise RuntimeError(
+                    f"DP bucket filled {b.ready}/{len(b.params)} times: a "
+                    "backward touched this module's params twice without "
+                    "finish(), or a cross-model backward leaked into it "
+                    "(freeze the other model's params — see "
+                    "Trainer._forward_g)")
             if b.work is not None:
                 b.work.wait()
             for p, v in zip(b.params, b.views):

@@ -84,3 +84,60 @@ def test_world1_noop():
     y = model(torch.randn(1, 16, 8))
     y.mean().backward()
     engine.finish()  # no-op, must not raise
+
+
+def _gan_worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from melgan_amd.config import HParam, merge_default
+        from melgan_amd.engine.trainer import Trainer
+
+        raw = """
+data: {train_dir: null, val_dir: null}
+train: {batch_size: 2, num_workers: 0, discriminator_train_start_steps: 0}
+audio: {segment_length: 4096, hop_length: 256}
+model: {out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}
+log: {chkpt_dir: /tmp/dpg_ck, log_dir: /tmp/dpg_lg}
+backend: {amp: none, fused_adam: false}
+"""
+        hp = merge_default(HParam(raw=raw))
+        torch.manual_seed(100 + rank)  # different init; broadcast equalises
+        tr = Trainer(hp, f"dpgan{rank}", rank=rank, world_size=world,
+                     device=torch.device("cpu"), enable_tb=False)
+        torch.manual_seed(500 + rank)  # different data per rank
+        mel = torch.randn(2, 80, 16)
+        audio = torch.rand(2, 1, 4096) * 1.6 - 0.8
+        for _ in range(2):
+            # full GAN step incl. discriminator phase: the g-phase
+            # adversarial backward must NOT leak into the disc DP
+            # buckets (engine raises if it does)
+            g_loss, parts, d_loss = tr.train_step(mel, audio)
+            assert d_loss is not None and torch.isfinite(d_loss)
+        gp = torch.cat([p.data.flatten()
+                        for p in tr.generator.parameters()])
+        dp = torch.cat([p.data.flatten()
+                        for p in tr.discriminator.parameters()])
+        results[rank] = (gp, dp)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_full_gan_step_world2():
+    """Both ranks stay bit-identical through full adversarial steps
+    (regression for the cross-model DP-hook leak: the g-phase backward
+    used to fill the discriminator's buckets with adversarial grads,
+    making the d-step all-reduce stale — engine now raises on it)."""
+    world = 2
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_gan_worker, args=(world, 29537, results), nprocs=world,
+             join=True)
+    g0, d0 = results[0]
+    g1, d1 = results[1]
+    assert torch.equal(g0, g1), "generator params diverged across ranks"
+    assert torch.equal(d0, d1), "discriminator params diverged across ranks"
