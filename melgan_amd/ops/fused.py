@@ -91,18 +91,23 @@ class FusedWNConv1dFn(torch.autograd.Function):
             gz = C.act_bwd(gy, y, act, slope)
         else:
             gz = gy
-        gx = gw = gb = None
+        gx = gv = gg = gb = None
         if ctx.needs_input_grad[0]:
             gx = C.conv1d_bwd_data(gz, w_c, stride, dilation, pad, reflect,
                                    groups, x.size(2))
             if act_before:
                 C.leaky_bwd_inplace(gx, x, slope)
-        gw_full = C.conv1d_bwd_weight(gz, x, stride, dilation, pad, reflect,
-                                      groups, w_c.size(2), act_before, slope)
-        gv, gg = _wn_backward(gw_full, v, g, n)
-        if ctx.has_bias:
+        if ctx.needs_input_grad[1]:  # frozen params (g-phase adversarial
+            # forward) skip the whole weight-grad GEMM
+            gw_full = C.conv1d_bwd_weight(gz, x, stride, dilation, pad,
+                                          reflect, groups, w_c.size(2),
+                                          act_before, slope)
+            gv, gg = _wn_backward(gw_full, v, g, n)
+            gv = gv.to(v.dtype)
+            gg = gg.to(g.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[3]:
             gb = C.conv1d_bias_grad(gz)
-        return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
+        return (gx, gv, gg, gb, None, None, None,
                 None, None, None, None, None, None, None)
 
 
@@ -144,19 +149,22 @@ class FusedWNConvT1dFn(torch.autograd.Function):
         x, v, g, n, w_c = ctx.saved_tensors
         stride, pad, out_pad, slope, act_before = ctx.geom
         gy = gy.contiguous()
-        gx = gb = None
+        gx = gv = gg = gb = None
         if ctx.needs_input_grad[0]:
             gx = C.conv_transpose1d_bwd_data(gy, w_c, stride, pad, out_pad,
                                              x.size(2))
             if act_before:
                 C.leaky_bwd_inplace(gx, x, slope)
-        gw_full = C.conv_transpose1d_bwd_weight(gy, x, stride, pad, out_pad,
-                                                w_c.size(2), act_before,
-                                                slope)
-        gv, gg = _wn_backward(gw_full, v, g, n)
-        if ctx.has_bias:
+        if ctx.needs_input_grad[1]:
+            gw_full = C.conv_transpose1d_bwd_weight(gy, x, stride, pad,
+                                                    out_pad, w_c.size(2),
+                                                    act_before, slope)
+            gv, gg = _wn_backward(gw_full, v, g, n)
+            gv = gv.to(v.dtype)
+            gg = gg.to(g.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[3]:
             gb = C.conv1d_bias_grad(gy)
-        return (gx, gv.to(v.dtype), gg.to(g.dtype), gb, None, None, None,
+        return (gx, gv, gg, gb, None, None, None,
                 None, None, None, None)
 
 
