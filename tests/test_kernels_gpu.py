@@ -702,3 +702,54 @@ def test_chunked_inference_gpu_bf16():
     # interior identical (bf16: boundary frames may differ by rounding)
     d = (chunked.float() - full.float()).abs().max()
     assert d < 1e-2, float(d)
+
+
+def test_conv_shape_fuzz_bf16():
+    """Seeded random-shape sweep over the supported conv envelope:
+    every (Cin, Cout, K, stride, dil, groups, pad_mode) drawn here must
+    match the fp32 composite within bf16 tolerance, fwd + backward."""
+    import random
+
+    rng = random.Random(1234)
+    from melgan_amd.ops.fused import fused_conv1d
+
+    for case in range(18):
+        groups = rng.choice([1, 1, 1, 2, 4])
+        cpg_in = rng.choice([1, 2, 4, 8, 16, 24, 40])
+        cpg_out = rng.choice([1, 2, 4, 8, 16, 24])
+        cin, cout = cpg_in * groups, cpg_out * groups
+        k = rng.choice([1, 3, 5, 7, 11, 15, 41])
+        stride = rng.choice([1, 1, 1, 2, 4])
+        dil = rng.choice([1, 1, 3, 9]) if stride == 1 else 1
+        pm = rng.choice(["zero", "reflect"]) if stride == 1 else "zero"
+        t = rng.choice([67, 128, 300, 1024])
+        span = (k - 1) * dil + 1
+        if pm == "reflect" and (k - 1) // 2 * dil >= t:
+            pm = "zero"
+        if t + (0 if pm == "none" else 2 * ((k - 1) // 2 * dil)) < span:
+            continue
+        mod = _mkconv(cin, cout, k, stride=stride, dilation=dil,
+                      groups=groups, pad_mode=pm, act=None)
+        torch.manual_seed(case)
+        x16 = torch.randn(2, cin, t, device=DEV).to(torch.bfloat16)
+        tag = f"fuzz#{case} {cin}x{cout}k{k}s{stride}d{dil}g{groups}{pm}t{t}"
+
+        x1 = x16.clone().requires_grad_(True)
+        y1 = fused_conv1d(mod, x1)
+        gy = torch.randn_like(y1)
+        y1.backward(gy)
+        gx16 = x1.grad.clone()
+        gv16 = mod.weight_v.grad.clone()
+        for p in mod.parameters():
+            p.grad = None
+
+        x2 = x16.float().clone().requires_grad_(True)
+        y2 = mod._composite(x2)
+        y2.backward(gy.float())
+        assert _rel(y1, y2) < 2e-2, f"{tag} y {_rel(y1, y2):.4f}"
+        assert _rel(gx16, x2.grad) < 3e-2, \
+            f"{tag} gx {_rel(gx16, x2.grad):.4f}"
+        assert _rel(gv16, mod.weight_v.grad) < 3e-2, \
+            f"{tag} gv {_rel(gv16, mod.weight_v.grad):.4f}"
+        for p in mod.parameters():
+            p.grad = None
