@@ -749,7 +749,13 @@ def test_conv_shape_fuzz_bf16():
         assert _rel(y1, y2) < 2e-2, f"{tag} y {_rel(y1, y2):.4f}"
         assert _rel(gx16, x2.grad) < 3e-2, \
             f"{tag} gx {_rel(gx16, x2.grad):.4f}"
-        assert _rel(gv16, mod.weight_v.grad) < 3e-2, \
-            f"{tag} gv {_rel(gv16, mod.weight_v.grad):.4f}"
+        if cpg_in * k > 1:
+            assert _rel(gv16, mod.weight_v.grad) < 3e-2, \
+                f"{tag} gv {_rel(gv16, mod.weight_v.grad):.4f}"
+        else:
+            # degenerate 1-element-per-channel v: grad_v == 0 exactly
+            # (weight norm leaves only the sign of v observable), so a
+            # relative comparison is roundoff-vs-roundoff; bound abs.
+            assert gv16.float().abs().max() < 1e-1, tag + " gv~0"
         for p in mod.parameters():
             p.grad = None
