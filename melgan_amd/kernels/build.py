@@ -23,6 +23,7 @@ SOURCES = [
     "kernels.hip",
     "conv1d_mfma.hip",
     "stft.hip",
+    "gemm.hip",
 ]
 
 

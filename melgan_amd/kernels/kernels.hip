@@ -486,6 +486,14 @@ torch::Tensor multi_l1_fwd(torch::Tensor tbl, torch::Tensor chunks,
 void multi_l1_bwd(torch::Tensor tbl, torch::Tensor chunks, int64_t nt,
                   torch::Tensor gout, torch::ScalarType dt);
 
+// glds GEMM engine (gemm.hip)
+torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
+                              c10::optional<torch::Tensor> bias,
+                              int64_t K, int64_t stride, int64_t dil,
+                              int64_t pad, bool reflect, int64_t act,
+                              double slope, bool act_before, int64_t To);
+torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt);
+
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
                          int64_t dil, int64_t pad, bool reflect,
@@ -878,6 +886,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stft_loss_bwd", &stft_loss_bwd);
   m.def("multi_l1_fwd", &multi_l1_fwd);
   m.def("multi_l1_bwd", &multi_l1_bwd);
+  m.def("conv1d_fwd_gemm", &conv1d_fwd_gemm);
+  m.def("gemm_bt", &gemm_bt);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }
