@@ -1295,3 +1295,15 @@ torch::Tensor conv_transpose1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
   return conv1d_fwd_mfma_generic(x_up, wf, bias_p, 1, 1, pad2, false, 1,
                                  act, slope, act_before, To);
 }
+
+// python-visible un-pad fold (GEMM bwd-data path): gxpad (B, C, Tpad)
+// -> gx (B, C, Tin) with reflect edge folding
+torch::Tensor unpad_fold_py(torch::Tensor gxpad, int64_t Tin, int64_t pad,
+                            bool reflect) {
+  const int B = gxpad.size(0), C = gxpad.size(1);
+  const int src_len = gxpad.size(2);
+  TORCH_CHECK(src_len >= Tin + 2 * pad);
+  auto gx = torch::empty({B, C, (int)Tin}, gxpad.options());
+  run_unpad_fold(gxpad, gx, B * C, (int)Tin, (int)pad, reflect, src_len);
+  return gx;
+}

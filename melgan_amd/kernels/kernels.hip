@@ -493,6 +493,8 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                               int64_t pad, bool reflect, int64_t act,
                               double slope, bool act_before, int64_t To);
 torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt);
+torch::Tensor unpad_fold_py(torch::Tensor gxpad, int64_t Tin, int64_t pad,
+                            bool reflect);
 
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
@@ -888,6 +890,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_l1_bwd", &multi_l1_bwd);
   m.def("conv1d_fwd_gemm", &conv1d_fwd_gemm);
   m.def("gemm_bt", &gemm_bt);
+  m.def("unpad_fold", &unpad_fold_py);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

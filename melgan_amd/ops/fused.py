@@ -69,6 +69,31 @@ def _gemm_fwd_eligible(x, cin, cout, k, stride, groups):
             and cin * k >= 384 and cout >= 64)
 
 
+def _gemm_bwd_eligible(gz_dtype, cin, cout, k):
+    import os as _os
+
+    if _os.environ.get("MELGAN_AMD_NO_GEMM", "0") == "1":
+        return False
+    return (gz_dtype == torch.bfloat16 and cout * k >= 384 and cin >= 64)
+
+
+def _apad_bwd_cached(mod, w_c: torch.Tensor):
+    """Flipped/permuted weights as the bwd-data GEMM A operand:
+    wf[ci][(co,k')] = w[co][ci][K-1-k'], zero-padded to ceil32."""
+    key = (w_c.data_ptr(), w_c.shape)
+    cache = getattr(mod, "_apad_bwd_cache", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    cout, cin, k = w_c.shape
+    nk = cout * k
+    kp = (nk + 31) // 32 * 32
+    wf = w_c.flip(-1).permute(1, 0, 2).reshape(cin, nk)
+    apad = torch.zeros(cin, kp, device=w_c.device, dtype=w_c.dtype)
+    apad[:, :nk] = wf
+    mod._apad_bwd_cache = (key, apad)
+    return apad
+
+
 def _apad_cached(mod, w_c: torch.Tensor):
     """Zero-padded (Cout, ceil32(Cin*K)) GEMM A operand, cached per fold."""
     key = (w_c.data_ptr(), w_c.shape)
@@ -94,8 +119,8 @@ def _wn_backward(gw: torch.Tensor, v: torch.Tensor, g: torch.Tensor,
 
 class FusedWNConv1dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, v, g, bias, w_c, n, apad, stride, dilation, pad,
-                reflect, groups, act, slope, act_before):
+    def forward(ctx, x, v, g, bias, w_c, n, apad, apad_bwd, stride,
+                dilation, pad, reflect, groups, act, slope, act_before):
         C = _ext.require_ext()
         b_c = bias.to(x.dtype) if bias is not None else None
         if apad is not None:
@@ -108,6 +133,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
             y = C.conv1d_fwd(x, w_c, b_c, stride, dilation, pad, reflect,
                              groups, act, slope, act_before)
         ctx.save_for_backward(x, v, g, n, w_c, y)
+        ctx.apad_bwd = apad_bwd
         ctx.geom = (stride, dilation, pad, reflect, groups, act, slope,
                     act_before)
         ctx.has_bias = bias is not None
@@ -126,8 +152,18 @@ class FusedWNConv1dFn(torch.autograd.Function):
             gz = gy
         gx = gv = gg = gb = None
         if ctx.needs_input_grad[0]:
-            gx = C.conv1d_bwd_data(gz, w_c, stride, dilation, pad, reflect,
-                                   groups, x.size(2))
+            if ctx.apad_bwd is not None:
+                # GEMM engine: gxpad = conv(gz, w-flipped), then un-pad
+                K = w_c.size(2)
+                Tpad = x.size(2) + 2 * pad
+                gxpad = C.conv1d_fwd_gemm(
+                    gz, ctx.apad_bwd, None, K, 1, dilation,
+                    (K - 1) * dilation, False, 0, 0.0, False, Tpad)
+                gx = C.unpad_fold(gxpad, x.size(2), pad, reflect) \
+                    if pad > 0 else gxpad
+            else:
+                gx = C.conv1d_bwd_data(gz, w_c, stride, dilation, pad,
+                                       reflect, groups, x.size(2))
             if act_before:
                 C.leaky_bwd_inplace(gx, x, slope)
         if ctx.needs_input_grad[1]:  # frozen params (g-phase adversarial
@@ -140,7 +176,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
             gg = gg.to(g.dtype)
         if ctx.has_bias and ctx.needs_input_grad[3]:
             gb = C.conv1d_bias_grad(gz)
-        return (gx, gv, gg, gb, None, None, None, None,
+        return (gx, gv, gg, gb, None, None, None, None, None,
                 None, None, None, None, None, None, None)
 
 
@@ -169,10 +205,14 @@ def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
                             mod.act_slope, mod.act_before)
     w_c, n = _wn_cached(mod, x.dtype)
     apad = _apad_cached(mod, w_c) if gemm_ok else None
+    apad_bwd = _apad_bwd_cached(mod, w_c) \
+        if (mod.stride == 1 and mod.groups == 1 and
+            _gemm_bwd_eligible(x.dtype, mod.in_channels, mod.out_channels,
+                               mod.kernel_size)) else None
     return FusedWNConv1dFn.apply(
-        x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, apad, mod.stride,
-        mod.dilation, mod.pad, mod.pad_mode == "reflect", mod.groups,
-        mod.act, mod.act_slope, mod.act_before)
+        x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, apad, apad_bwd,
+        mod.stride, mod.dilation, mod.pad, mod.pad_mode == "reflect",
+        mod.groups, mod.act, mod.act_slope, mod.act_before)
 
 
 class FusedWNConvT1dFn(torch.autograd.Function):
