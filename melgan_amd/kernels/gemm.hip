@@ -86,18 +86,8 @@ gemm_bt_glds_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
   // [2 buffers][A tile | B tile], each tile 128*32 ushorts
   const int TILE = GBM * GBK;
 
-  // XCD-aware block swizzle (guide T1, bijective variant): contiguous
-  // grid chunks per XCD so neighbouring tiles share L2-resident panels
-  const int nwg = gridDim.x * gridDim.y;
-  const int orig = blockIdx.y * gridDim.x + blockIdx.x;
-  const int q = nwg / 8, rr = nwg % 8;
-  const int xcd = orig % 8, pos = orig / 8;
-  const int swzid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) +
-                     (xcd - rr) * q) + pos;
-  const int bidx = swzid % gridDim.x;
-  const int bidy = swzid / gridDim.x;
-  const int m0 = bidy * GBM;
-  const int64_t n0 = (int64_t)bidx * GBN;
+  const int m0 = blockIdx.y * GBM;
+  const int64_t n0 = (int64_t)blockIdx.x * GBN;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
