@@ -184,6 +184,119 @@ __global__ void im2col_bt_kernel(const unsigned short* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// split-K fp32 variant for the weight gradient:
+//   out[M][N] += A[M][Kred] * Bt[N][Kred]^T   (fp32 atomic epilogue)
+// A = gz^T (Cout x B*To), Bt = k-major im2col (Cin*K x B*To); Kred =
+// B*To is huge, so grid.z slices the K loop to fill the chip.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+gemm_bt_splitk_f32_kernel(const unsigned short* __restrict__ A,
+                          const unsigned short* __restrict__ Bt,
+                          float* __restrict__ out, int M, int64_t N,
+                          int64_t Kred, int nsplit) {
+  extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
+  const int TILE = GBM * GBK;
+
+  const int m0 = blockIdx.y * GBM;
+  const int64_t n0 = (int64_t)blockIdx.x * GBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int row = lane & 15;
+  const int kk0 = (lane >> 4) * 8;
+
+  f32x4g acc[4][4] = {};
+
+  const int ksteps = (int)(Kred / GBK);
+  const int per = (ksteps + nsplit - 1) / nsplit;
+  const int ks_lo = blockIdx.z * per;
+  const int ks_hi = min(ks_lo + per, ksteps);
+  if (ks_lo >= ks_hi) return;
+
+  stage_tile_glds(A, m0, M, Kred, ks_lo * GBK, smem, wid, lane);
+  stage_tile_glds(Bt, n0, N, Kred, ks_lo * GBK, smem + TILE, wid, lane);
+  __syncthreads();
+
+  int cur = 0;
+  for (int ks = ks_lo; ks < ks_hi; ++ks) {
+    if (ks + 1 < ks_hi) {
+      unsigned short* nb = smem + (cur ^ 1) * 2 * TILE;
+      stage_tile_glds(A, m0, M, Kred, (ks + 1) * GBK, nb, wid, lane);
+      stage_tile_glds(Bt, n0, N, Kred, (ks + 1) * GBK, nb + TILE, wid,
+                      lane);
+    }
+    const unsigned short* la = smem + cur * 2 * TILE;
+    const unsigned short* lb = la + TILE;
+    bf16x8g af[4], bf[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int r = wr * 64 + m * 16 + row;
+      af[m] = *(const bf16x8g*)(la + (r * 64 + swz(r, kk0 * 2)) / 2);
+    }
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int r = wc * 64 + n * 16 + row;
+      bf[n] = *(const bf16x8g*)(lb + (r * 64 + swz(r, kk0 * 2)) / 2);
+    }
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bf[n], acc[m][n], 0, 0, 0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int col = lane & 15;
+  const int row4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int64_t gn = n0 + wc * 64 + n * 16 + col;
+      if (gn >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gm = m0 + wr * 64 + m * 16 + row4 + r;
+        if (gm >= M) continue;
+        atomicAdd(&out[(int64_t)gm * N + gn], acc[m][n][r]);
+      }
+    }
+}
+
+// k-major im2col for the weight-grad GEMM:
+//   out[ci*K + k][b*To + t] = pre(x[b][ci][t*stride + k*dil - pad])
+// (bt contiguous per row: both reads and writes coalesce)
+__global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
+                                 unsigned short* __restrict__ out,
+                                 int Bb, int Cin, int Tin, int To, int K,
+                                 int stride, int dil, int pad, bool reflect,
+                                 int NKp, bool act_before, float slope) {
+  const int64_t NB = (int64_t)Bb * To;
+  const int64_t total = (int64_t)NKp * NB;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  const int NK = Cin * K;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += step) {
+    const int64_t bt = i % NB;
+    const int kk = (int)(i / NB);
+    unsigned short v = 0;
+    if (kk < NK) {
+      const int t = (int)(bt % To);
+      const int b = (int)(bt / To);
+      const int ci = kk / K;
+      const int k = kk - ci * K;
+      const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+      v = load_pad_bf16(xr, t * stride + k * dil - pad, Tin, reflect);
+      if (act_before) v = leaky_bf16(v, slope);
+    }
+    out[i] = v;
+  }
+}
+
 }  // namespace melgan
 
 using namespace melgan;
@@ -249,4 +362,45 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt) {
                      (unsigned short*)y.data_ptr(), M, N, Kp, (int)N, 0,
                      0.f);
   return y.view({M, N});
+}
+
+// gw[Cout][Cin][K] (fp32) via the split-K GEMM engine.
+torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
+                                     int64_t K, int64_t stride, int64_t dil,
+                                     int64_t pad, bool reflect,
+                                     bool act_before, double slope) {
+  const int Bb = gz.size(0), Cout = gz.size(1), To = gz.size(2);
+  const int Cin = x.size(1), Tin = x.size(2);
+  const int64_t Kred = (int64_t)Bb * To;
+  TORCH_CHECK(Kred % GBK == 0, "B*To must be a multiple of 32");
+  const int NK = Cin * (int)K;
+  const int NKp = (NK + 31) / 32 * 32;  // N padded only for tiling
+  auto stream = at::hip::getCurrentHIPStream();
+
+  auto gzT = gz.permute({1, 0, 2}).reshape({Cout, Kred}).contiguous();
+  auto btk = torch::empty({(int64_t)NKp, Kred}, x.options());
+  {
+    const int64_t n = (int64_t)NKp * Kred;
+    hipLaunchKernelGGL(im2col_tk_kernel,
+                       dim3((int)std::min<int64_t>(4096, (n + 255) / 256)),
+                       dim3(256), 0, stream,
+                       (const unsigned short*)x.data_ptr(),
+                       (unsigned short*)btk.data_ptr(), Bb, Cin, Tin, To,
+                       (int)K, (int)stride, (int)dil, (int)pad, reflect,
+                       NKp, act_before, (float)slope);
+  }
+  auto gw = torch::zeros({Cout, (int64_t)NKp},
+                         x.options().dtype(torch::kFloat));
+  const int mt = gcdiv(Cout, GBM), nt = gcdiv(NKp, GBN);
+  const int ksteps = (int)(Kred / GBK);
+  int nsplit = (int)std::min<int64_t>(
+      ksteps, std::max<int64_t>(1, 512 / std::max(1, mt * nt)));
+  const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);
+  dim3 grid(nt, mt, nsplit);
+  hipLaunchKernelGGL(gemm_bt_splitk_f32_kernel, grid, dim3(256), lds,
+                     stream, (const unsigned short*)gzT.data_ptr(),
+                     (const unsigned short*)btk.data_ptr(),
+                     gw.data_ptr<float>(), Cout, (int64_t)NKp, Kred,
+                     nsplit);
+  return gw.narrow(1, 0, NK).reshape({Cout, Cin, (int64_t)K}).contiguous();
 }

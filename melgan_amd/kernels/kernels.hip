@@ -495,6 +495,10 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
 torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt);
 torch::Tensor unpad_fold_py(torch::Tensor gxpad, int64_t Tin, int64_t pad,
                             bool reflect);
+torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
+                                     int64_t K, int64_t stride, int64_t dil,
+                                     int64_t pad, bool reflect,
+                                     bool act_before, double slope);
 
 torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias, int64_t stride,
@@ -891,6 +895,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_fwd_gemm", &conv1d_fwd_gemm);
   m.def("gemm_bt", &gemm_bt);
   m.def("unpad_fold", &unpad_fold_py);
+  m.def("conv1d_bwd_weight_gemm", &conv1d_bwd_weight_gemm);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

@@ -168,9 +168,20 @@ class FusedWNConv1dFn(torch.autograd.Function):
                 C.leaky_bwd_inplace(gx, x, slope)
         if ctx.needs_input_grad[1]:  # frozen params (g-phase adversarial
             # forward) skip the whole weight-grad GEMM
-            gw_full = C.conv1d_bwd_weight(gz, x, stride, dilation, pad,
-                                          reflect, groups, w_c.size(2),
-                                          act_before, slope)
+            K = w_c.size(2)
+            import os as _os
+
+            if (gz.dtype == torch.bfloat16 and stride == 1 and groups == 1
+                    and x.size(1) * K >= 384 and gz.size(1) >= 64
+                    and (gz.size(0) * gz.size(2)) % 32 == 0
+                    and _os.environ.get("MELGAN_AMD_NO_GEMM", "0") != "1"):
+                gw_full = C.conv1d_bwd_weight_gemm(
+                    gz, x, K, stride, dilation, pad, reflect, act_before,
+                    slope)
+            else:
+                gw_full = C.conv1d_bwd_weight(gz, x, stride, dilation, pad,
+                                              reflect, groups, K,
+                                              act_before, slope)
             gv, gg = _wn_backward(gw_full, v, g, n)
             gv = gv.to(v.dtype)
             gg = gg.to(g.dtype)
