@@ -200,7 +200,13 @@ def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
                                  mod.kernel_size, mod.stride, mod.groups)
     if mod._removed:
         C = _ext.require_ext()
-        w_c = mod.weight.to(x.dtype)
+        wk = (x.dtype, mod.weight._version)
+        wc_cache = getattr(mod, "_wcast_cache", None)
+        if wc_cache is not None and wc_cache[0] == wk:
+            w_c = wc_cache[1]
+        else:
+            w_c = mod.weight.to(x.dtype).contiguous()
+            mod._wcast_cache = (wk, w_c)
         b_c = mod.bias.to(x.dtype) if mod.bias is not None else None
         if gemm_ok:
             apad = _apad_cached(mod, w_c)
@@ -270,7 +276,13 @@ def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
     x = x.contiguous()
     if mod._removed:
         C = _ext.require_ext()
-        w_c = mod.weight.to(x.dtype)
+        wk = (x.dtype, mod.weight._version)
+        wc_cache = getattr(mod, "_wcast_cache", None)
+        if wc_cache is not None and wc_cache[0] == wk:
+            w_c = wc_cache[1]
+        else:
+            w_c = mod.weight.to(x.dtype).contiguous()
+            mod._wcast_cache = (wk, w_c)
         b_c = mod.bias.to(x.dtype) if mod.bias is not None else None
         return C.conv_transpose1d_fwd(x, w_c, b_c, mod.stride, mod.padding,
                                       mod.output_padding, mod.act_slope,
