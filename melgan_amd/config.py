@@ -84,7 +84,6 @@ class HParam(Dotdict):
 
 def merge_default(hp: HParam) -> HParam:
     """Fill defaults for keys the reference family treats as optional."""
-    d = hp.setdefault
     # Groups
     for g in ("data", "train", "audio", "model", "log", "backend"):
         if hp.get(g) is None:
