@@ -30,6 +30,23 @@ from melgan_amd.engine.trainer import Trainer
 from melgan_amd.parallel.dp import init_distributed
 
 
+def _reexec_torchrun(nproc: int):
+    """Replace this process with `torch.distributed.run --nproc-per-node N
+    bench.py <same args>` (rendezvous on 127.0.0.1 — container hostnames
+    may not resolve)."""
+    import socket
+    import sys
+
+    with socket.socket() as s:  # pick a free rendezvous port
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    os.execvp(sys.executable, [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
+        "--master-port", str(port),
+        os.path.abspath(__file__), *sys.argv[1:]])
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -43,14 +60,30 @@ def main():
                     help="include discriminator phase (full GAN step)")
     ap.add_argument("--no-adversarial", dest="adversarial",
                     action="store_false")
-    ap.add_argument("--inference-rtf", action="store_true",
-                    help="also measure inference RTF @22.05 kHz")
+    ap.add_argument("--no-rtf", dest="rtf", action="store_false",
+                    default=True,
+                    help="skip the inference-RTF measurement")
+    ap.add_argument("--inference-rtf", dest="rtf", action="store_true",
+                    help="(default) also measure inference RTF")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.gpus > 1 and world == 1:
+        # invoked directly with --gpus N: re-exec under torchrun so the
+        # bench REALLY runs N ranks (one per GPU) instead of silently
+        # measuring one
+        _reexec_torchrun(args.gpus)
+        return  # unreachable (execvp replaces the process)
     if world > 1:
         init_distributed()
+        got = torch.distributed.get_world_size()
+        if args.gpus > 1 and got != args.gpus:
+            raise SystemExit(
+                f"bench.py: asked for --gpus {args.gpus} but world_size is "
+                f"{got} — launcher/env mismatch, refusing to report a bogus "
+                "n_gpus")
+        world = got
     have_gpu = torch.cuda.is_available()
     device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0"))) \
         if have_gpu else torch.device("cpu")
@@ -109,8 +142,15 @@ def main():
     peak_hbm_gb = round(torch.cuda.max_memory_allocated() / 2 ** 30, 2) \
         if have_gpu else None
 
+    # DP overlap artifact: how much of the gradient
+    # all-reduce was already complete when backward ended, per model
+    comm = None
+    if world > 1 and trainer.dp_g.last_stats:
+        comm = {"g": trainer.dp_g.last_stats,
+                "d": trainer.dp_d.last_stats or None}
+
     rtf = None
-    if args.inference_rtf and rank == 0:
+    if args.rtf and rank == 0:
         rtf = measure_rtf(hp, device)
 
     if rank == 0:
@@ -144,6 +184,8 @@ def main():
         if rtf is not None:
             out["config"][
                 f"inference_rtf_{hp.audio.sampling_rate // 1000}k"] = rtf
+        if comm is not None:
+            out["config"]["dp_overlap"] = comm
         print(json.dumps(out))
 
 
@@ -153,6 +195,9 @@ def measure_rtf(hp, device, n_frames: int = 860, reps: int = 20,
     (bf16 weights/activations; PQMF synthesis fp32)."""
     from melgan_amd.models import build_generator
     from melgan_amd.ops.pqmf import PQMF
+
+    if device.type != "cuda":  # CPU smoke runs: keep it to seconds
+        n_frames, reps = 128, 2
 
     g = build_generator(hp).to(device).eval()
     g.remove_weight_norm()

@@ -153,9 +153,14 @@ def create_dataloader(hp, train: bool = True, synthetic: bool = False,
     if world_size > 1:
         sampler = torch.utils.data.distributed.DistributedSampler(
             dataset, num_replicas=world_size, rank=rank, shuffle=train)
+    # validation items from disk are FULL-LENGTH uncropped clips of varying
+    # size — default_collate can't batch them, so val loaders use batch 1
+    # (the family convention); synthetic data is fixed-shape and batches fine
+    batch_size = hp.train.batch_size \
+        if (train or isinstance(dataset, SyntheticMelDataset)) else 1
     return DataLoader(
         dataset,
-        batch_size=hp.train.batch_size,
+        batch_size=batch_size,
         shuffle=(sampler is None and train),
         sampler=sampler,
         num_workers=hp.train.num_workers,

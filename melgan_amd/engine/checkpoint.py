@@ -32,7 +32,13 @@ def save_checkpoint(path_dir: str, name: str, step: int, generator,
 
 
 def load_checkpoint(path: str, map_location="cpu") -> dict:
-    return torch.load(path, map_location=map_location, weights_only=False)
+    # the saved dict is plain tensors/ints/str, so the safe loader works;
+    # fall back only for legacy pickles (e.g. foreign optimizer states)
+    try:
+        return torch.load(path, map_location=map_location, weights_only=True)
+    except Exception:
+        return torch.load(path, map_location=map_location,
+                          weights_only=False)
 
 
 def load_flexible(module: torch.nn.Module, state: dict, strict_first=True,

@@ -129,19 +129,8 @@ class Trainer:
             self.discriminator, self.optim_g, self.optim_d, self.hp.hp_str)
 
     # ------------------------------------------------------------------
-    def _forward_g(self, mel, audio):
-        """Generator forward + all generator losses. Returns (loss, parts,
-        fake full-band audio)."""
-        hp = self.hp
-        with torch.autocast("cuda", dtype=self.amp_dtype,
-                            enabled=self.use_amp):
-            fake_sub = self.generator(mel)  # (B, N_bands, S/N)
-        fake_sub = fake_sub.float()
-        if self.multiband:
-            fake = self.pqmf.synthesis(fake_sub)  # (B, 1, S)
-        else:
-            fake = fake_sub
-
+    def _stft_losses(self, fake_sub, fake, audio):
+        """Multi-resolution STFT losses (full-band + sub-band mix)."""
         sc_full, mag_full = self.stft_full(fake.squeeze(1),
                                            audio.squeeze(1).float())
         stft_loss = sc_full + mag_full
@@ -152,10 +141,44 @@ class Trainer:
             sc_sub, mag_sub = self.stft_sub(fake_sub, audio_sub)
             sub_loss = sc_sub + mag_sub
             stft_loss = 0.5 * stft_loss + 0.5 * sub_loss
+        return stft_loss, sub_loss
 
-        g_loss = stft_loss
-        adv = feat = None
-        if self.step >= hp.train.discriminator_train_start_steps:
+    def _forward_g(self, mel, audio):
+        """Generator forward + generator losses (no discriminator reuse —
+        kept for the warmup phase and standalone callers). Returns
+        (loss, parts, fake full-band audio)."""
+        with torch.autocast("cuda", dtype=self.amp_dtype,
+                            enabled=self.use_amp):
+            fake_sub = self.generator(mel)  # (B, N_bands, S/N)
+        fake_sub = fake_sub.float()
+        fake = self.pqmf.synthesis(fake_sub) if self.multiband else fake_sub
+        stft_loss, sub_loss = self._stft_losses(fake_sub, fake, audio)
+        return stft_loss, dict(stft=stft_loss, sub=sub_loss, adv=None,
+                               feat=None), fake
+
+    def train_step(self, mel, audio):
+        hp = self.hp
+        mel = mel.to(self.device, non_blocking=True)
+        audio = audio.to(self.device, non_blocking=True)
+        adversarial = self.step >= hp.train.discriminator_train_start_steps
+
+        tm = self.timer
+        # ---- generator forward + losses --------------------------------
+        tm.mark("g_fwd")
+        self.dp_g.zero_grad(self.optim_g)
+        g_loss, parts, fake = self._forward_g(mel, audio)
+
+        # ---- discriminator forwards (both phases share D(real)) --------
+        d_loss = None
+        if adversarial:
+            tm.mark("d_fwd")
+            self.dp_d.zero_grad(self.optim_d)
+            with torch.autocast("cuda", dtype=self.amp_dtype,
+                                enabled=self.use_amp):
+                d_real = self.discriminator(audio.float())
+                d_fake_det = self.discriminator(fake.detach())
+            d_loss = L.discriminator_loss(d_real, d_fake_det)
+
             # Freeze discriminator PARAMS for the generator-phase
             # adversarial forward: the g loss needs gradients THROUGH the
             # discriminator to `fake`, not INTO its weights.  Without
@@ -163,33 +186,24 @@ class Trainer:
             # kernels and (b) fires the disc's DP bucket hooks with
             # g-phase gradients, which would make the d-step's
             # all-reduce ship STALE gradients at world_size>1.
+            tm.mark("g_adv_fwd")
             for p in self.discriminator.parameters():
                 p.requires_grad_(False)
             try:
                 with torch.autocast("cuda", dtype=self.amp_dtype,
                                     enabled=self.use_amp):
                     d_fake = self.discriminator(fake)
-                    with torch.no_grad():
-                        d_real = self.discriminator(audio.float())
             finally:
                 for p in self.discriminator.parameters():
                     p.requires_grad_(True)
             adv = L.adversarial_g_loss(d_fake)
+            # feature-matching targets reuse the d-phase D(real) features
+            # (detached inside the loss) — D runs on real audio ONCE/step
             feat = L.feature_matching_loss(d_fake, d_real)
             g_loss = g_loss + self.lambda_adv * adv + self.lambda_feat * feat
-        return g_loss, dict(stft=stft_loss, sub=sub_loss, adv=adv,
-                            feat=feat), fake
+            parts["adv"], parts["feat"] = adv, feat
 
-    def train_step(self, mel, audio):
-        hp = self.hp
-        mel = mel.to(self.device, non_blocking=True)
-        audio = audio.to(self.device, non_blocking=True)
-
-        tm = self.timer
-        # ---- generator ------------------------------------------------
-        tm.mark("g_fwd")
-        self.optim_g.zero_grad(set_to_none=True)
-        g_loss, parts, fake = self._forward_g(mel, audio)
+        # ---- generator update (D still at pre-step weights) -------------
         tm.mark("g_bwd")
         g_loss.backward()
         tm.mark("g_comm_wait")
@@ -197,16 +211,8 @@ class Trainer:
         tm.mark("g_opt")
         self.optim_g.step()
 
-        # ---- discriminator --------------------------------------------
-        d_loss = None
-        if self.step >= hp.train.discriminator_train_start_steps:
-            tm.mark("d_fwd")
-            self.optim_d.zero_grad(set_to_none=True)
-            with torch.autocast("cuda", dtype=self.amp_dtype,
-                                enabled=self.use_amp):
-                d_real = self.discriminator(audio.float())
-                d_fake = self.discriminator(fake.detach())
-            d_loss = L.discriminator_loss(d_real, d_fake)
+        # ---- discriminator update ---------------------------------------
+        if adversarial:
             tm.mark("d_bwd")
             d_loss.backward()
             tm.mark("d_comm_wait")
@@ -229,40 +235,55 @@ class Trainer:
                             hp.audio.segment_length // hp.audio.hop_length)
         t0 = time.time()
         steps_t0 = self.step
-        data_iter = itertools.cycle(trainloader) \
-            if hasattr(trainloader, "__iter__") else trainloader
-        for mel, audio in data_iter:
+        sampler = getattr(trainloader, "sampler", None)
+        # explicit epoch loop (NOT itertools.cycle, which would memoise the
+        # first epoch's random crops forever and pin their memory, and never
+        # reshuffle DistributedSampler shards)
+        for epoch in itertools.count():
             if self.step >= max_steps:
                 break
-            g_loss, parts, d_loss = self.train_step(mel, audio)
+            if isinstance(sampler,
+                          torch.utils.data.distributed.DistributedSampler):
+                sampler.set_epoch(epoch)
+            saw_batch = False
+            for mel, audio in trainloader:
+                saw_batch = True
+                if self.step >= max_steps:
+                    break
+                g_loss, parts, d_loss = self.train_step(mel, audio)
 
-            if self.step % hp.log.summary_interval == 0 and self.rank == 0:
-                if self.device.type == "cuda":
-                    torch.cuda.synchronize()
-                dt = time.time() - t0
-                nsteps = self.step - steps_t0
-                mfps = (nsteps * frames_per_batch * self.world / dt) \
-                    if dt > 0 else 0.0
-                self.writer.log_training(
-                    self.step, g_loss=g_loss, d_loss=d_loss,
-                    stft=parts["stft"], sub=parts["sub"], adv=parts["adv"],
-                    feat=parts["feat"], mel_frames_per_sec=mfps,
-                    **{f"ms_{k}": v for k, v in self.last_phases.items()})
-                if self.logger:
-                    self.logger.info(
-                        f"step {self.step} g_loss {float(g_loss):.4f}"
-                        + (f" d_loss {float(d_loss):.4f}" if d_loss is not None
-                           else "")
-                        + f" | {mfps:,.0f} mel-frames/s")
-                t0 = time.time()
-                steps_t0 = self.step
+                if self.step % hp.log.summary_interval == 0 and \
+                        self.rank == 0:
+                    if self.device.type == "cuda":
+                        torch.cuda.synchronize()
+                    dt = time.time() - t0
+                    nsteps = self.step - steps_t0
+                    mfps = (nsteps * frames_per_batch * self.world / dt) \
+                        if dt > 0 else 0.0
+                    self.writer.log_training(
+                        self.step, g_loss=g_loss, d_loss=d_loss,
+                        stft=parts["stft"], sub=parts["sub"],
+                        adv=parts["adv"], feat=parts["feat"],
+                        mel_frames_per_sec=mfps,
+                        **{f"ms_{k}": v
+                           for k, v in self.last_phases.items()})
+                    if self.logger:
+                        self.logger.info(
+                            f"step {self.step} g_loss {float(g_loss):.4f}"
+                            + (f" d_loss {float(d_loss):.4f}"
+                               if d_loss is not None else "")
+                            + f" | {mfps:,.0f} mel-frames/s")
+                    t0 = time.time()
+                    steps_t0 = self.step
 
-            if valloader is not None and \
-                    self.step % hp.log.validation_interval == 0:
-                self.validate(valloader)
+                if valloader is not None and \
+                        self.step % hp.log.validation_interval == 0:
+                    self.validate(valloader)
 
-            if self.step % hp.log.save_interval == 0:
-                self.save()
+                if self.step % hp.log.save_interval == 0:
+                    self.save()
+            if not saw_batch:  # exhausted one-shot iterable: stop
+                break
         return self.step
 
     @torch.no_grad()

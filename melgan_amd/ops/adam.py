@@ -52,11 +52,24 @@ class FusedAdam(torch.optim.Optimizer):
             e = _ext.ext()
             if e is not None and params[0].is_cuda and hasattr(
                     e, "multi_tensor_adam_tbl"):
-                tbl, chunks, nt = self._tables(
-                    id(group), params, grads, exp_avgs, exp_avg_sqs)
-                e.multi_tensor_adam_tbl(
-                    tbl, chunks, nt, group["lr"], beta1, beta2,
-                    group["eps"], group["weight_decay"], steps[0])
+                # the fused kernel takes ONE bias-correction step count per
+                # launch — partition by step value so a param whose grad
+                # first appeared later (param added / temporarily grad-less)
+                # gets its own correct correction.  In steady state all
+                # params share one step -> one launch, cached tables.
+                by_step: dict[int, list[int]] = {}
+                for i, s in enumerate(steps):
+                    by_step.setdefault(s, []).append(i)
+                for s, idxs in by_step.items():
+                    sel = (lambda lst: [lst[i] for i in idxs])
+                    # cache key from the PARTITION (not the step value,
+                    # which increments every step and would kill caching)
+                    tbl, chunks, nt = self._tables(
+                        (id(group), tuple(idxs)), sel(params), sel(grads),
+                        sel(exp_avgs), sel(exp_avg_sqs))
+                    e.multi_tensor_adam_tbl(
+                        tbl, chunks, nt, group["lr"], beta1, beta2,
+                        group["eps"], group["weight_decay"], s)
                 # the kernel writes params through raw pointers, which
                 # does NOT bump torch's _version counter — bump an
                 # explicit epoch so weight-norm fold caches (keyed on it)

@@ -10,11 +10,18 @@ engine designed for the MI355X node topology:
   the whole generator fits in ~2): total gradient volume is tiny
   (<=40 MB fp32/step) so ring-all-reduce LATENCY, not bandwidth,
   dominates on xGMI — few launches beat many
+- the bucket views ARE the gradient storage: `zero_grad()` binds each
+  `p.grad` to its slice of the flat bucket, so backward accumulates
+  straight into the bucket and `finish()` has nothing to copy (the
+  round-1 version copied grad->bucket and averaged-bucket->grad — two
+  full-gradient passes per step on the latency-critical path)
 - each bucket's all-reduce is launched asynchronously the moment its
   last gradient is accumulated (post-accumulate-grad hooks), so comm
   runs on RCCL's comm stream concurrently with the REST of backward
-- `finish()` waits the handles and un-flattens averaged grads before
-  optimizer.step()
+- `finish()` waits the handles before optimizer.step() and records a
+  per-bucket (ready -> done) timeline so multi-GPU runs emit an
+  overlap artifact (how much comm was already complete when backward
+  ended) instead of just a throughput number
 
 CPU CI runs the same engine over the gloo backend (world_size>1 multi-
 process tests), so the distributed path is covered without
@@ -22,6 +29,8 @@ a GPU.
 """
 
 from __future__ import annotations
+
+import time
 
 import torch
 import torch.distributed as dist
@@ -57,6 +66,8 @@ class GradBucket:
             off += p.numel()
         self.ready = 0
         self.work = None
+        self.t_ready = 0.0
+        self.t_done = 0.0
 
     def reset(self):
         self.ready = 0
@@ -68,13 +79,14 @@ class BucketAllReduceDP:
 
     Usage:
         engine = BucketAllReduceDP(model, bucket_cap_mb=25)
-        loss.backward()          # hooks launch async all-reduces
-        engine.finish()          # wait + write averaged grads back
+        engine.zero_grad(optimizer)   # zero buckets + bind p.grad -> views
+        loss.backward()               # hooks launch async all-reduces
+        engine.finish()               # wait handles (grads already in place)
         optimizer.step()
     """
 
     def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 25.0,
-                 process_group=None, grad_dtype: torch.dtype = torch.float32):
+                 process_group=None):
         self.module = module
         self.group = process_group
         self.world = dist.get_world_size(process_group) \
@@ -83,6 +95,7 @@ class BucketAllReduceDP:
         self._hooks = []
         self.buckets: list[GradBucket] = []
         self._param_bucket: dict[int, tuple[GradBucket, int]] = {}
+        self.last_stats: dict = {}
         if not self.enabled:
             return
 
@@ -98,7 +111,9 @@ class BucketAllReduceDP:
         groups: list[list[torch.nn.Parameter]] = []
         for p in reversed(params):
             sz = p.numel() * p.element_size()
-            if cur and cur_bytes + sz > cap:
+            # p.grad will BE a view of the flat bucket, so a bucket must be
+            # dtype-homogeneous
+            if cur and (cur_bytes + sz > cap or p.dtype != cur[0].dtype):
                 groups.append(cur)
                 cur, cur_bytes = [], 0
             cur.append(p)
@@ -107,10 +122,12 @@ class BucketAllReduceDP:
             groups.append(cur)
         device = params[0].device
         for g in groups:
-            b = GradBucket(g, device, grad_dtype)
+            # grad storage shares the param dtype so p.grad can BE the view
+            b = GradBucket(g, device, g[0].dtype)
             self.buckets.append(b)
             for i, p in enumerate(g):
                 self._param_bucket[id(p)] = (b, i)
+        self._bind_grads()
 
         for p in params:
             h = p.register_post_accumulate_grad_hook(self._on_grad)
@@ -124,19 +141,54 @@ class BucketAllReduceDP:
             if isinstance(t, torch.Tensor) and t.numel() > 0:
                 dist.broadcast(t.data, src=0, group=self.group)
 
+    def _bind_grads(self):
+        for b in self.buckets:
+            for p, v in zip(b.params, b.views):
+                p.grad = v
+
+    def zero_grad(self, optimizer=None):
+        """Zero gradients for the wrapped module's next step.
+
+        Enabled: zero the flat buckets and (re)bind p.grad to the bucket
+        views so backward accumulates in place — pointer-stable across
+        steps (FusedAdam's cached device tables and hipGraph capture both
+        rely on that).  Disabled (world 1): plain set_to_none semantics
+        via the optimizer if given, else per-param.
+        """
+        if not self.enabled:
+            if optimizer is not None:
+                optimizer.zero_grad(set_to_none=True)
+            else:
+                for p in self.module.parameters():
+                    p.grad = None
+            return
+        for b in self.buckets:
+            b.flat.zero_()
+            b.reset()
+        self._bind_grads()
+
     def _on_grad(self, p: torch.nn.Parameter):
         b, i = self._param_bucket[id(p)]
-        b.views[i].copy_(p.grad.detach().to(b.flat.dtype))
+        if p.grad is not None and p.grad.data_ptr() != b.views[i].data_ptr():
+            # a backward ran without zero_grad() (autograd allocated fresh
+            # storage) — fold it into the bucket and rebind
+            b.views[i].add_(p.grad.detach().to(b.flat.dtype))
+            p.grad = b.views[i]
         b.ready += 1
         if b.ready == len(b.params):
             b.flat.div_(self.world)
+            b.t_ready = time.perf_counter()
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                      group=self.group, async_op=True)
 
     def finish(self):
-        """Wait outstanding all-reduces; write averaged grads back."""
+        """Wait outstanding all-reduces (averaged grads are already in
+        p.grad — the views are the storage). Records overlap stats."""
         if not self.enabled:
             return
+        t_bwd_end = time.perf_counter()
+        launched = [b for b in self.buckets if b.work is not None]
+        pre_done = 0
         for b in self.buckets:
             if b.ready not in (0, len(b.params)):
                 raise RuntimeError(
@@ -146,10 +198,22 @@ class BucketAllReduceDP:
                     "(freeze the other model's params — see "
                     "Trainer._forward_g)")
             if b.work is not None:
+                if b.work.is_completed():
+                    pre_done += 1
                 b.work.wait()
-            for p, v in zip(b.params, b.views):
-                if p.grad is not None:
-                    p.grad.detach().copy_(v.to(p.grad.dtype))
+                b.t_done = time.perf_counter()
+        t_end = time.perf_counter()
+        if launched:
+            self.last_stats = {
+                "buckets": len(launched),
+                "bytes": sum(b.numel * b.flat.element_size()
+                             for b in launched),
+                "pre_completed": pre_done,
+                "finish_wait_ms": round((t_end - t_bwd_end) * 1e3, 3),
+                "bucket_ms": [round((b.t_done - b.t_ready) * 1e3, 3)
+                              for b in launched],
+            }
+        for b in self.buckets:
             b.reset()
 
     def comm_bytes_per_step(self) -> int:
