@@ -232,6 +232,151 @@ def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
         mod.groups, mod.act, mod.act_slope, mod.act_before)
 
 
+# ---------------------------------------------------------------------------
+# ConvTranspose1d on the glds GEMM engine via phase decomposition
+# (VERDICT r1 "missing #5": kill the (s-1)/s zero-stuff waste).
+#
+# Identity: with k = 2r, stride r, pad p (To = T*r exactly for the
+# generator's geometry, even and odd r):
+#     y[co][u*r + phi] = sum_{ci,i} w[ci][co][e(phi) + i*r]
+#                                   * x[ci][u + c(phi) - i]
+# where e = (phi+p) % r, c = (phi+p) // r in {0,1}, i in {0,1}.  That is
+# a stride-1 K=3 pad=1 conv of x with M = r*Cout output rows (m = phi*
+# Cout + co) whose A operand re-folds w; the result un-interleaves with
+# one permute.  Backward-data is a stride-r K=2r conv of gy with
+# A = w.reshape(Cin, Cout*K); backward-weight is the split-K weight-grad
+# GEMM with (gz=leaky(x), x=gy) swapped roles.
+# ---------------------------------------------------------------------------
+def _convt_gemm_ok(x, mod):
+    import os as _os
+
+    if _os.environ.get("MELGAN_AMD_NO_GEMM", "0") == "1":
+        return False
+    return (x.dtype == torch.bfloat16
+            and mod.out_channels * mod.stride >= 64
+            and mod.in_channels >= 64)
+
+
+def _aphase_cached(mod, w_c: torch.Tensor):
+    """Phase-folded fwd A operand: (r*Cout, ceil32(3*Cin)) bf16,
+    A[phi*Cout+co][ci*3+jj] = w[ci][co][e + (1+c-jj)*r] (0 if i out of
+    range).  Cached per weight fold."""
+    key = (w_c.data_ptr(), w_c.shape)
+    cache = getattr(mod, "_aphase_cache", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    cin, cout, K = w_c.shape
+    r, p = mod.stride, mod.padding
+    idx = getattr(mod, "_aphase_idx", None)
+    if idx is None or idx[0].device != w_c.device:
+        import torch as _t
+
+        kidx = _t.full((r, 3), 0, dtype=_t.long)
+        mask = _t.zeros(r, 3, dtype=_t.bool)
+        for phi in range(r):
+            e = (phi + p) % r
+            c = (phi + p) // r
+            for jj in range(3):
+                i = 1 + c - jj
+                if 0 <= i < 2 and e + i * r < K:
+                    kidx[phi, jj] = e + i * r
+                    mask[phi, jj] = True
+        idx = (kidx.to(w_c.device), mask.to(w_c.device))
+        mod._aphase_idx = idx
+    kidx, mask = idx
+    kp = (3 * cin + 31) // 32 * 32
+    # (cout, cin, r, 3) gather -> (r, cout, cin*3)
+    sel = w_c.permute(1, 0, 2)[:, :, kidx.reshape(-1)] \
+        .reshape(cout, cin, r, 3)
+    sel = sel * mask.reshape(1, 1, r, 3).to(w_c.dtype)
+    A = torch.zeros(r, cout, kp, device=w_c.device, dtype=w_c.dtype)
+    A[:, :, :3 * cin] = sel.permute(2, 0, 1, 3).reshape(r, cout, cin * 3)
+    A = A.reshape(r * cout, kp)
+    mod._aphase_cache = (key, A)
+    return A
+
+
+def _a2_cached(mod, w_c: torch.Tensor):
+    """Bwd-data A operand: (Cin, ceil32(Cout*K)) — gx is a stride-r conv
+    of gy with the untouched weight block as rows."""
+    key = (w_c.data_ptr(), w_c.shape)
+    cache = getattr(mod, "_a2_cache", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    cin, cout, K = w_c.shape
+    nk = cout * K
+    kp = (nk + 31) // 32 * 32
+    A = torch.zeros(cin, kp, device=w_c.device, dtype=w_c.dtype)
+    A[:, :nk] = w_c.reshape(cin, nk)
+    mod._a2_cache = (key, A)
+    return A
+
+
+def _biasph_cached(mod):
+    """bias repeated per phase (float), m = phi*Cout + co ordering."""
+    b = mod.bias
+    key = (b._version, getattr(b, "_wn_epoch", 0))
+    cache = getattr(mod, "_biasph_cache", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    bp = b.detach().float().repeat(mod.stride).contiguous()
+    mod._biasph_cache = (key, bp)
+    return bp
+
+
+def _convt_gemm_fwd(C, x, aph, bias_ph, cout, stride, slope, act_before):
+    B, _, T = x.shape
+    yph = C.conv1d_fwd_gemm(x, aph, bias_ph, 3, 1, 1, 1, False,
+                            ACT_NONE, slope, act_before, T)
+    return yph.view(B, stride, cout, T).permute(0, 2, 3, 1) \
+        .reshape(B, cout, T * stride)
+
+
+class ConvTGemmFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, v, g, bias, w_c, n, aph, bias_ph, a2, stride, pad,
+                out_pad, slope, act_before):
+        C = _ext.require_ext()
+        y = _convt_gemm_fwd(C, x, aph, bias_ph, w_c.size(1), stride, slope,
+                            act_before)
+        ctx.save_for_backward(x, v, g, n, w_c, a2)
+        ctx.geom = (stride, pad, out_pad, slope, act_before)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        x, v, g, n, w_c, a2 = ctx.saved_tensors
+        stride, pad, out_pad, slope, act_before = ctx.geom
+        gy = gy.contiguous()
+        K = w_c.size(2)
+        T = x.size(2)
+        gx = gv = gg = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = C.conv1d_fwd_gemm(gy, a2, None, K, stride, 1, pad, False,
+                                   ACT_NONE, 0.0, False, T)
+            if act_before:
+                C.leaky_bwd_inplace(gx, x, slope)
+        if ctx.needs_input_grad[1]:
+            lx = torch.nn.functional.leaky_relu(x, slope) \
+                if act_before else x
+            if (x.size(0) * T) % 32 == 0 and gy.dtype == torch.bfloat16:
+                gw_full = C.conv1d_bwd_weight_gemm(
+                    lx.contiguous(), gy, K, stride, 1, pad, False, False,
+                    0.0)
+            else:
+                gw_full = C.conv_transpose1d_bwd_weight(
+                    gy, x, stride, pad, out_pad, K, act_before, slope)
+            gv, gg = _wn_backward(gw_full, v, g, n)
+            gv = gv.to(v.dtype)
+            gg = gg.to(g.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[3]:
+            gb = C.conv1d_bias_grad(gy)
+        return (gx, gv, gg, gb, None, None, None, None, None, None, None,
+                None, None, None)
+
+
 class FusedWNConvT1dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, v, g, bias, w_c, n, stride, pad, out_pad, slope,
@@ -274,6 +419,7 @@ def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
     if torch.is_autocast_enabled("cuda") and x.dtype == torch.float32:
         x = x.to(torch.get_autocast_dtype("cuda"))
     x = x.contiguous()
+    gemm_ok = _convt_gemm_ok(x, mod)
     if mod._removed:
         C = _ext.require_ext()
         wk = (x.dtype, mod.weight._version)
@@ -283,11 +429,25 @@ def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
         else:
             w_c = mod.weight.to(x.dtype).contiguous()
             mod._wcast_cache = (wk, w_c)
+        if gemm_ok:
+            aph = _aphase_cached(mod, w_c)
+            bias_ph = _biasph_cached(mod) if mod.bias is not None else None
+            return _convt_gemm_fwd(C, x, aph, bias_ph, mod.out_channels,
+                                   mod.stride, mod.act_slope,
+                                   mod.act_before)
         b_c = mod.bias.to(x.dtype) if mod.bias is not None else None
         return C.conv_transpose1d_fwd(x, w_c, b_c, mod.stride, mod.padding,
                                       mod.output_padding, mod.act_slope,
                                       mod.act_before)
     w_c, n = _wn_cached(mod, x.dtype)
+    if gemm_ok:
+        aph = _aphase_cached(mod, w_c)
+        a2 = _a2_cached(mod, w_c)
+        bias_ph = _biasph_cached(mod) if mod.bias is not None else None
+        return ConvTGemmFn.apply(
+            x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, aph, bias_ph,
+            a2, mod.stride, mod.padding, mod.output_padding, mod.act_slope,
+            mod.act_before)
     return FusedWNConvT1dFn.apply(
         x, mod.weight_v, mod.weight_g, mod.bias, w_c, n, mod.stride,
         mod.padding, mod.output_padding, mod.act_slope, mod.act_before)

@@ -1,6 +1,7 @@
 """CPU-testable units of the GPU dispatch layer: eligibility predicates
 and table builders (pure logic, no kernels)."""
 
+import pytest
 import torch
 
 from melgan_amd.ops import fused
@@ -52,3 +53,66 @@ def test_phase_timer_disabled_cpu():
     tm.mark("x")
     tm.mark(None)
     assert tm.collect() == {}
+
+
+# ---------------------------------------------------------------------------
+# ConvT phase-decomposed GEMM route (fused._aphase_cached et al.): verify
+# the A-fold identity ON CPU by emulating the GEMM engine with F.conv1d.
+# ---------------------------------------------------------------------------
+def _convt_mod(cin, cout, r):
+    from melgan_amd.ops.wnconv import WNConvTranspose1d
+
+    torch.manual_seed(0)
+    return WNConvTranspose1d(cin, cout, r)
+
+
+@pytest.mark.parametrize("r", [2, 3, 4, 5, 8])
+def test_convt_aphase_fold_matches_composite(r):
+    import torch.nn.functional as F
+
+    mod = _convt_mod(12, 6, r)
+    w_c = mod.current_weight().detach()
+    aph = fused._aphase_cached(mod, w_c)
+    bias_ph = fused._biasph_cached(mod)
+    cin, cout, K = w_c.shape
+    B, T = 2, 17
+    x = torch.randn(B, cin, T)
+    # emulate conv1d_fwd_gemm: K=3 pad=1 conv, act_before leaky prologue
+    xp = F.leaky_relu(x, mod.act_slope) if mod.act_before else x
+    Aw = aph[:, :3 * cin].reshape(r * cout, cin, 3)
+    yph = F.conv1d(xp, Aw, bias_ph, padding=1)
+    y = yph.view(B, r, cout, T).permute(0, 2, 3, 1).reshape(B, cout, T * r)
+    ref = mod._composite(x)
+    assert (y - ref).abs().max() < 1e-5
+
+
+@pytest.mark.parametrize("r", [2, 4, 8])
+def test_convt_a2_fold_matches_bwd_data(r):
+    import torch.nn.functional as F
+
+    mod = _convt_mod(10, 5, r)
+    mod.act_before = False
+    w_c = mod.current_weight().detach()
+    a2 = fused._a2_cached(mod, w_c)
+    cin, cout, K = w_c.shape
+    B, T = 2, 13
+    x = torch.randn(B, cin, T, requires_grad=True)
+    y = F.conv_transpose1d(x, w_c, None, stride=r, padding=mod.padding,
+                           output_padding=mod.output_padding)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    # emulate conv1d_fwd_gemm(gy, a2, K=2r, stride=r, pad=p)
+    Aw = a2[:, :cout * K].reshape(cin, cout, K)
+    gx = F.conv1d(gy, Aw, None, stride=r, padding=mod.padding)
+    assert (gx - x.grad).abs().max() < 1e-5
+
+
+def test_convt_gemm_eligibility():
+    mod8 = _convt_mod(512, 256, 8)
+    mod_pqmf = _convt_mod(4, 1, 4)  # PQMF synthesis shape: stays off GEMM
+    bf = torch.zeros(1, 512, 8, dtype=torch.bfloat16)
+    f32 = torch.zeros(1, 512, 8)
+    assert fused._convt_gemm_ok(bf, mod8)
+    assert not fused._convt_gemm_ok(f32, mod8)
+    bf_p = torch.zeros(1, 4, 8, dtype=torch.bfloat16)
+    assert not fused._convt_gemm_ok(bf_p, mod_pqmf)

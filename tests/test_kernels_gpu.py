@@ -500,6 +500,45 @@ def test_conv_transpose_mfma_bf16(ratio, cin, cout, t):
         f"{tag} gv rel {_rel(gv16, mod.weight_v.grad):.4f}"
 
 
+@pytest.mark.parametrize("ratio,cin,cout,t", [
+    (8, 512, 256, 64), (4, 256, 128, 512), (5, 96, 48, 160)])
+def test_conv_transpose_gemm_route_act_before(ratio, cin, cout, t):
+    """The generator's real upsample config (LeakyReLU prologue) through
+    the phase-decomposed GEMM route — fwd + both backwards vs fp32
+    composite."""
+    from melgan_amd.ops.wnconv import WNConvTranspose1d
+    from melgan_amd.ops.fused import fused_conv_transpose1d, _convt_gemm_ok
+
+    torch.manual_seed(3)
+    mod = WNConvTranspose1d(cin, cout, ratio, act_before=True).to(DEV)
+    x16 = torch.randn(2, cin, t, device=DEV).to(torch.bfloat16)
+    assert _convt_gemm_ok(x16, mod)  # the GEMM route must be the one taken
+
+    x1 = x16.clone().requires_grad_(True)
+    y1 = fused_conv_transpose1d(mod, x1)
+    assert y1.size(2) == t * ratio
+    gy = torch.randn_like(y1)
+    y1.backward(gy)
+    gx16 = x1.grad.clone()
+    gv16 = mod.weight_v.grad.clone()
+    gg16 = mod.weight_g.grad.clone()
+    gb16 = mod.bias.grad.clone()
+    for p in mod.parameters():
+        p.grad = None
+
+    x2 = x16.float().clone().requires_grad_(True)
+    y2 = mod._composite(x2)
+    y2.backward(gy.float())
+    tag = f"convT gemm r{ratio} act_before"
+    assert _rel(y1, y2) < 2e-2, tag + " y"
+    assert _rel(gx16, x2.grad) < 3e-2, \
+        f"{tag} gx rel {_rel(gx16, x2.grad):.4f}"
+    assert _rel(gv16, mod.weight_v.grad) < 3e-2, \
+        f"{tag} gv rel {_rel(gv16, mod.weight_v.grad):.4f}"
+    assert _rel(gg16, mod.weight_g.grad) < 3e-2, tag + " gg"
+    assert _rel(gb16, mod.bias.grad) < 3e-2, tag + " gb"
+
+
 # ---------------------------------------------------------------------------
 # Fused STFT pipeline (frame+window prologue, |.| epilogue around rocFFT)
 # ---------------------------------------------------------------------------
