@@ -24,6 +24,7 @@ SOURCES = [
     "conv1d_mfma.hip",
     "stft.hip",
     "gemm.hip",
+    "xstage.hip",
 ]
 
 

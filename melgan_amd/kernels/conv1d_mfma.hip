@@ -988,6 +988,28 @@ conv1d_bwd_weight_mfma_kernel(const unsigned short* __restrict__ gz,
 // ===========================================================================
 using namespace melgan;
 
+// xstage tier for the grouped strided k=41 towers (xstage.hip)
+bool xstage_fwd_eligible(int64_t Cin, int64_t Cout, int64_t K,
+                         int64_t stride, int64_t dil, int64_t groups,
+                         bool reflect);
+torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
+                         const float* bias_p, int64_t stride, int64_t pad,
+                         int64_t groups, int64_t act, double slope,
+                         bool act_before, int64_t To);
+bool xstage_bwd_data_eligible(int64_t Cin, int64_t Cout, int64_t K,
+                              int64_t stride, int64_t dil, int64_t groups,
+                              bool reflect);
+torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
+                              int64_t stride, int64_t pad, int64_t groups,
+                              int64_t Tin);
+bool xstage_bwd_weight_eligible(int64_t Cin, int64_t Cout, int64_t K,
+                                int64_t stride, int64_t dil,
+                                int64_t groups, bool reflect);
+torch::Tensor xstage_bwd_weight(torch::Tensor gz, torch::Tensor x,
+                                int64_t stride, int64_t pad,
+                                int64_t groups, int64_t K, bool act_b,
+                                double slope);
+
 static inline int cdiv_i(int64_t a, int64_t b) {
   return (int)((a + b - 1) / b);
 }
@@ -1102,6 +1124,9 @@ torch::Tensor conv1d_fwd_mfma_generic(torch::Tensor x, torch::Tensor w,
   const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
   const int Cout = w.size(0), K = w.size(2);
   const int Copg = Cout / (int)groups;
+  if (xstage_fwd_eligible(Cin, Cout, K, stride, dil, groups, reflect))
+    return xstage_fwd(x, w, bias_p, stride, pad, groups, act, slope,
+                      act_before, To);
   auto y = torch::empty({B, Cout, (int)To}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
   if (Copg <= 16) {
@@ -1184,6 +1209,8 @@ torch::Tensor conv1d_bwd_data_mfma_generic(torch::Tensor gz,
   const int pad2 = (K - 1) * (int)dil;
   const int Tpad = (int)Tin + 2 * (int)pad;
   const int CpgX = Cpg;  // (per-group input channels == gx channels/group)
+  if (xstage_bwd_data_eligible(Cin, Cout, K, stride, dil, groups, reflect))
+    return xstage_bwd_data(gz, w, stride, pad, groups, Tin);
   if (!reflect && stride > 1 && dil == 1 &&
       CpgX * stride <= 64) {  // (cig,phi) rows per group fit few m-tiles
     auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
@@ -1234,6 +1261,10 @@ torch::Tensor conv1d_bwd_weight_mfma(torch::Tensor gz, torch::Tensor x,
   const int Cpg = Cin / (int)groups;
   const int Copg = Cout / (int)groups;
   const int N = Cpg * (int)K;
+  if (!act_a &&
+      xstage_bwd_weight_eligible(Cin, Cout, K, stride, dil, groups,
+                                 reflect))
+    return xstage_bwd_weight(gz, x, stride, pad, groups, K, act_b, slope);
   auto gw = torch::zeros({Cout, Cpg, (int)K},
                          x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
