@@ -158,29 +158,41 @@ gemm_bt_glds_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
 
 // im2col to the B^T image: out[(b*To + t)][ci*K + k] =
 //   pre(x[b][ci][t*stride + k*dil - pad]); zero for kk >= Cin*K (K-pad).
+// Vectorized: one 16-B store (8 kk) per thread; the (ci,k) split is
+// computed once and carried, not divided per element.
 __global__ void im2col_bt_kernel(const unsigned short* __restrict__ x,
                                  unsigned short* __restrict__ out,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int Kp, bool act_before, float slope) {
-  const int64_t total = (int64_t)Bb * To * Kp;
+  const int K8 = Kp >> 3;
+  const int64_t total = (int64_t)Bb * To * K8;
   const int64_t step = (int64_t)gridDim.x * blockDim.x;
   const int NK = Cin * K;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total; i += step) {
-    const int kk = (int)(i % Kp);
-    const int64_t bt = i / Kp;
-    unsigned short v = 0;
-    if (kk < NK) {
-      const int t = (int)(bt % To);
-      const int b = (int)(bt / To);
-      const int ci = kk / K;
-      const int k = kk - ci * K;
-      const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-      v = load_pad_bf16(xr, t * stride + k * dil - pad, Tin, reflect);
-      if (act_before) v = leaky_bf16(v, slope);
+    const int k8 = (int)(i % K8);
+    const int64_t bt = i / K8;
+    const int t = (int)(bt % To);
+    const int b = (int)(bt / To);
+    int kk = k8 * 8;
+    int ci = kk / K;
+    int k = kk - ci * K;
+    const int tpos = t * stride - pad;
+    union { unsigned short e[8]; ushort4 q[2]; } v = {};
+#pragma unroll
+    for (int j = 0; j < 8; ++j, ++kk, ++k) {
+      if (k >= K) { k = 0; ++ci; }
+      if (kk < NK) {
+        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+        unsigned short e = load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
+        if (act_before) e = leaky_bf16(e, slope);
+        v.e[j] = e;
+      }
     }
-    out[i] = v;
+    ushort4* dst = (ushort4*)(out + bt * Kp + (int64_t)k8 * 8);
+    dst[0] = v.q[0];
+    dst[1] = v.q[1];
   }
 }
 
@@ -269,31 +281,42 @@ gemm_bt_splitk_f32_kernel(const unsigned short* __restrict__ A,
 
 // k-major im2col for the weight-grad GEMM:
 //   out[ci*K + k][b*To + t] = pre(x[b][ci][t*stride + k*dil - pad])
-// (bt contiguous per row: both reads and writes coalesce)
+// Vectorized: one 16-B store (8 bt) per thread; b/t carried, not
+// divided per element.  (B*To is a multiple of 32 — checked by caller.)
 __global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
                                  unsigned short* __restrict__ out,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int NKp, bool act_before, float slope) {
   const int64_t NB = (int64_t)Bb * To;
-  const int64_t total = (int64_t)NKp * NB;
+  const int64_t NB8 = NB >> 3;
+  const int64_t total = (int64_t)NKp * NB8;
   const int64_t step = (int64_t)gridDim.x * blockDim.x;
   const int NK = Cin * K;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total; i += step) {
-    const int64_t bt = i % NB;
-    const int kk = (int)(i / NB);
-    unsigned short v = 0;
+    const int64_t bt8 = i % NB8;
+    const int kk = (int)(i / NB8);
+    union { unsigned short e[8]; ushort4 q[2]; } v = {};
     if (kk < NK) {
-      const int t = (int)(bt % To);
-      const int b = (int)(bt / To);
       const int ci = kk / K;
       const int k = kk - ci * K;
-      const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-      v = load_pad_bf16(xr, t * stride + k * dil - pad, Tin, reflect);
-      if (act_before) v = leaky_bf16(v, slope);
+      int b = (int)(bt8 * 8 / To);
+      int t = (int)(bt8 * 8 - (int64_t)b * To);
+      const int koff = k * dil - pad;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+        unsigned short e = load_pad_bf16(xr, t * stride + koff, Tin,
+                                         reflect);
+        if (act_before) e = leaky_bf16(e, slope);
+        v.e[j] = e;
+        if (++t == To) { t = 0; ++b; }
+      }
     }
-    out[i] = v;
+    ushort4* dst = (ushort4*)(out + kk * NB + bt8 * 8);
+    dst[0] = v.q[0];
+    dst[1] = v.q[1];
   }
 }
 

@@ -191,13 +191,16 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
                        const unsigned short* __restrict__ w,
                        unsigned short* __restrict__ gx, int B, int Cin,
                        int Tin, int Cout, int To, int K, int stride,
-                       int pad, int groups, int nk, int NCH, int V) {
-  constexpr int LDR = 20;  // [t][co] row stride (16 co + 4 pad)
+                       int pad, int groups, int nk, int NCH, int V,
+                       int coplog) {
+  constexpr int LDR = 20;  // [t][co] row stride (<=16 co + pad)
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   unsigned short* lds_w = smem;                // [NCH][16][XLDW]
   unsigned short* lds_gz = smem + NCH * 16 * XLDW;  // [V][LDR]
 
   const int Cpg = Cin / groups;
+  const int CoP = 1 << coplog;       // Copg (power of 2, 4..16)
+  const int ipc = 32 >> coplog;      // i-values per 32-chunk
   const int g = blockIdx.y;
   const int u0 = blockIdx.x * 256;
   const int b = blockIdx.z;
@@ -215,23 +218,23 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
     const int kk = idx & 31;
     const int cig = r / stride;
     const int phi = r - cig * stride;
-    const int i = c * 2 + (kk >> 4);
-    const int co = kk & 15;
+    const int i = c * ipc + (kk >> coplog);
+    const int co = kk & (CoP - 1);
     const int k = phi + i * stride;
     unsigned short v = 0;
     if (i < nk && k < K)
-      v = w[((int64_t)(g * 16 + co) * Cpg + cig) * K + k];
+      v = w[((int64_t)(g * CoP + co) * Cpg + cig) * K + k];
     lds_w[(c * 16 + r) * XLDW + kk] = v;
   }
   // ---- stage gz tile transposed [t][co] (coalesced reads per co) ----
   const int tlo = u0 - nk + 1;
-  for (int idx = tid; idx < 16 * V; idx += 256) {
+  for (int idx = tid; idx < CoP * V; idx += 256) {
     const int co = idx / V;
     const int v = idx - co * V;
     const int t = tlo + v;
     unsigned short e = 0;
     if (t >= 0 && t < To)
-      e = gz[((int64_t)b * Cout + g * 16 + co) * To + t];
+      e = gz[((int64_t)b * Cout + g * CoP + co) * To + t];
     lds_gz[v * LDR + co] = e;
   }
   __syncthreads();
@@ -240,15 +243,18 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int colb = wid * 64;
   for (int c = 0; c < NCH; ++c) {
     bf16x8x a0 = *(const bf16x8x*)(lds_w + (c * 16 + row) * XLDW + kk0);
-    const int i = c * 2 + (kk0 >> 4);
-    const int co0 = kk0 & 15;
+    // fragment halves may straddle an i boundary when CoP == 4;
+    // i >= nk columns have zero A — clamp so the LDS read stays in range
+    const int i0 = min(c * ipc + (kk0 >> coplog), nk - 1);
+    const int co0 = kk0 & (CoP - 1);
+    const int i1 = min(c * ipc + ((kk0 + 4) >> coplog), nk - 1);
+    const int co1 = (kk0 + 4) & (CoP - 1);
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      const int vtx = colb + nf * 16 + row + nk - 1 - i;
-      const unsigned short* p = lds_gz + vtx * LDR + co0;
+      const int vt = colb + nf * 16 + row + nk - 1;
       bf16x8x bfr;
-      ((uint2*)&bfr)[0] = *(const uint2*)p;
-      ((uint2*)&bfr)[1] = *(const uint2*)(p + 4);
+      ((uint2*)&bfr)[0] = *(const uint2*)(lds_gz + (vt - i0) * LDR + co0);
+      ((uint2*)&bfr)[1] = *(const uint2*)(lds_gz + (vt - i1) * LDR + co1);
       acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfr, acc[nf],
                                                         0, 0, 0);
     }
@@ -283,7 +289,8 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
 __global__ void __launch_bounds__(256)
 xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
                          const unsigned short* __restrict__ x,
-                         float* __restrict__ gw, int B, int Cin, int Tin,
+                         float* __restrict__ gws,  // [nsplit][Cout][Cpg][K]
+                         int B, int Cin, int Tin,
                          int Cout, int To, int K, int stride, int pad,
                          int groups, float slope, bool act_b, int KP,
                          int nsplit) {
@@ -369,8 +376,11 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
     __syncthreads();
   }
 
+  // epilogue: plain stores into this z-slice's private slab (summed on
+  // the host) — z-sliced atomics on the tiny gw serialised ~256-way
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
+  float* slab = gws + (int64_t)blockIdx.z * Cout * Cpg * K;
 #pragma unroll
   for (int nf = 0; nf < NFRAG; ++nf) {
     const int i = wid * NFRAG + nf;
@@ -382,7 +392,7 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
     for (int r = 0; r < 4; ++r) {
       const int co = co_base + row4 + r;
       if (co < co_end)
-        atomicAdd(&gw[((int64_t)co * Cpg + cig) * K + k], acc[nf][r]);
+        slab[((int64_t)co * Cpg + cig) * K + k] = acc[nf][r];
     }
   }
 }
@@ -450,7 +460,9 @@ bool xstage_bwd_data_eligible(int64_t Cin, int64_t Cout, int64_t K,
   if (groups <= 1 || dil != 1 || reflect) return false;
   const int Cpg = (int)(Cin / groups);
   const int Copg = (int)(Cout / groups);
-  return Cpg * stride == 16 && Copg == 16 && K <= 48 && K >= 9;
+  const bool pow2 = (Copg & (Copg - 1)) == 0;
+  return Cpg * stride == 16 && pow2 && Copg >= 4 && Copg <= 16 &&
+         K <= 48 && K >= 9;
 }
 
 torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
@@ -459,8 +471,11 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cpg = w.size(1), K = w.size(2);
   const int Cin = Cpg * (int)groups;
+  const int Copg = Cout / (int)groups;
+  const int coplog = __builtin_ctz(Copg);
+  const int ipc = 32 / Copg;
   const int nk = (K + (int)stride - 1) / (int)stride;
-  const int NCH = (nk + 1) / 2;
+  const int NCH = (nk + ipc - 1) / ipc;
   const int V = (256 + nk + 7) / 8 * 8 + 8;
   const size_t lds = (size_t)(NCH * 16 * XLDW + V * 20) * 2;
   auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
@@ -472,7 +487,7 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
                      (const unsigned short*)w.data_ptr(),
                      (unsigned short*)gx.data_ptr(), B, Cin, (int)Tin,
                      Cout, To, K, (int)stride, (int)pad, (int)groups, nk,
-                     NCH, V);
+                     NCH, V, coplog);
   return gx;
 }
 
@@ -498,21 +513,23 @@ torch::Tensor xstage_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int Cpg = Cin / (int)groups;
   const int Copg = Cout / (int)groups;
   const int KP = (int)((K + 15) / 16 * 16);
-  auto gw = torch::zeros({Cout, Cpg, (int)K},
-                         x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   const int gy = (int)groups * xcdiv(Copg, 16);
   const int tchunks = xcdiv(To, 256);
   const int64_t total = (int64_t)B * tchunks;
   int nsplit = (int)std::min<int64_t>(
-      total, std::max<int64_t>(1, 1024 / std::max(1, gy)));
+      total, std::max<int64_t>(1, 768 / std::max(1, gy)));
+  // per-z private slabs, plain stores; summed below (every (z,co,cig,
+  // k<K) element is written by exactly one lane, so empty is safe)
+  auto gws = torch::empty({nsplit, Cout, Cpg, (int)K},
+                          x.options().dtype(torch::kFloat));
   const size_t lds = (size_t)(16 * (256 + 8) + 16 * 280) * 2;
   dim3 grid(1, gy, nsplit);
   hipLaunchKernelGGL(xstage_bwd_weight_kernel, grid, dim3(256), lds,
                      stream, (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
-                     gw.data_ptr<float>(), B, Cin, Tin, Cout, To, (int)K,
+                     gws.data_ptr<float>(), B, Cin, Tin, Cout, To, (int)K,
                      (int)stride, (int)pad, (int)groups, (float)slope,
                      act_b, KP, nsplit);
-  return gw;
+  return nsplit > 1 ? gws.sum(0) : gws.view({Cout, Cpg, (int)K});
 }

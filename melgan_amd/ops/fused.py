@@ -146,11 +146,19 @@ class FusedWNConv1dFn(torch.autograd.Function):
         stride, dilation, pad, reflect, groups, act, slope, act_before = \
             ctx.geom
         gy = gy.contiguous()
+        gb = None
+        want_gb = ctx.has_bias and ctx.needs_input_grad[3]
         if act != ACT_NONE and not act_before:
-            gz = C.act_bwd(gy, y, act, slope)
+            if want_gb and hasattr(C, "act_bwd_biasgrad"):
+                # one pass: gz written once, per-channel sum in the same
+                # kernel (saves the full gz re-read of a separate
+                # bias_grad launch)
+                gz, gb = C.act_bwd_biasgrad(gy, y, act, slope)
+            else:
+                gz = C.act_bwd(gy, y, act, slope)
         else:
             gz = gy
-        gx = gv = gg = gb = None
+        gx = gv = gg = None
         if ctx.needs_input_grad[0]:
             if ctx.apad_bwd is not None:
                 # GEMM engine: gxpad = conv(gz, w-flipped), then un-pad
@@ -185,7 +193,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
             gv, gg = _wn_backward(gw_full, v, g, n)
             gv = gv.to(v.dtype)
             gg = gg.to(g.dtype)
-        if ctx.has_bias and ctx.needs_input_grad[3]:
+        if want_gb and gb is None:
             gb = C.conv1d_bias_grad(gz)
         return (gx, gv, gg, gb, None, None, None, None, None,
                 None, None, None, None, None, None, None)
