@@ -47,27 +47,27 @@ DEV_INLINE unsigned funhalf(unsigned a, unsigned b) {
   return (a >> 16) | (b << 16);  // v_alignbit: elements (a.hi, b.lo)
 }
 
-// extract bf16 elements [d, d+8) from the 16 staged at 16B-aligned p;
-// du MUST be wave-uniform (readfirstlane'd) so the switch is scalar.
-DEV_INLINE bf16x8x extract8(const unsigned short* p, int du) {
-  const uint4 lo = *(const uint4*)p;
-  const uint4 hi = *(const uint4*)(p + 8);
-  const unsigned w0 = lo.x, w1 = lo.y, w2 = lo.z, w3 = lo.w;
-  const unsigned w4 = hi.x, w5 = hi.y, w6 = hi.z, w7 = hi.w;
+// extract bf16 elements [du, du+8) (du in [0, 10)) from 24 staged at
+// 16B-aligned p (three b128 reads); du MUST be wave-uniform
+// (readfirstlane'd) so the switch lowers to scalar branches.
+DEV_INLINE bf16x8x extract8w(const unsigned* w, int du) {
   unsigned o0, o1, o2, o3;
   switch (du) {
-    case 0: o0 = w0; o1 = w1; o2 = w2; o3 = w3; break;
-    case 1: o0 = funhalf(w0, w1); o1 = funhalf(w1, w2);
-            o2 = funhalf(w2, w3); o3 = funhalf(w3, w4); break;
-    case 2: o0 = w1; o1 = w2; o2 = w3; o3 = w4; break;
-    case 3: o0 = funhalf(w1, w2); o1 = funhalf(w2, w3);
-            o2 = funhalf(w3, w4); o3 = funhalf(w4, w5); break;
-    case 4: o0 = w2; o1 = w3; o2 = w4; o3 = w5; break;
-    case 5: o0 = funhalf(w2, w3); o1 = funhalf(w3, w4);
-            o2 = funhalf(w4, w5); o3 = funhalf(w5, w6); break;
-    case 6: o0 = w3; o1 = w4; o2 = w5; o3 = w6; break;
-    default: o0 = funhalf(w3, w4); o1 = funhalf(w4, w5);
-             o2 = funhalf(w5, w6); o3 = funhalf(w6, w7); break;
+    case 0: o0 = w[0]; o1 = w[1]; o2 = w[2]; o3 = w[3]; break;
+    case 1: o0 = funhalf(w[0], w[1]); o1 = funhalf(w[1], w[2]);
+            o2 = funhalf(w[2], w[3]); o3 = funhalf(w[3], w[4]); break;
+    case 2: o0 = w[1]; o1 = w[2]; o2 = w[3]; o3 = w[4]; break;
+    case 3: o0 = funhalf(w[1], w[2]); o1 = funhalf(w[2], w[3]);
+            o2 = funhalf(w[3], w[4]); o3 = funhalf(w[4], w[5]); break;
+    case 4: o0 = w[2]; o1 = w[3]; o2 = w[4]; o3 = w[5]; break;
+    case 5: o0 = funhalf(w[2], w[3]); o1 = funhalf(w[3], w[4]);
+            o2 = funhalf(w[4], w[5]); o3 = funhalf(w[5], w[6]); break;
+    case 6: o0 = w[3]; o1 = w[4]; o2 = w[5]; o3 = w[6]; break;
+    case 7: o0 = funhalf(w[3], w[4]); o1 = funhalf(w[4], w[5]);
+            o2 = funhalf(w[5], w[6]); o3 = funhalf(w[6], w[7]); break;
+    case 8: o0 = w[4]; o1 = w[5]; o2 = w[6]; o3 = w[7]; break;
+    default: o0 = funhalf(w[4], w[5]); o1 = funhalf(w[5], w[6]);
+             o2 = funhalf(w[6], w[7]); o3 = funhalf(w[7], w[8]); break;
   }
   union { unsigned u[4]; bf16x8x v; } r;
   r.u[0] = o0; r.u[1] = o1; r.u[2] = o2; r.u[3] = o3;
@@ -93,7 +93,7 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
                   int SP) {
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   unsigned short* lds_w = smem;                  // [NCH][16][XLDW]
-  unsigned short* lds_x = smem + NCH * 16 * XLDW;  // [Cpg][SP]
+  unsigned short* lds_x = smem + 6 * 16 * XLDW;  // [Cpg][SP]
 
   const int Copg = Cout / groups;
   const int Cpg = Cin / groups;
@@ -102,7 +102,6 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   const int mt = blockIdx.y - g * mtiles;
   const int co_base = g * Copg + mt * 16;
   const int co_end = (g + 1) * Copg;
-  const int t0 = blockIdx.x * 256;
   const int b = blockIdx.z;
   const int NSUB = Cpg * KT8;
 
@@ -112,7 +111,8 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   const int row = lane & 15;
   const int kk0 = (lane >> 4) * 8;
 
-  // ---- stage A (weights), mapped to the subcolumn order ----
+  // ---- stage A (weights) ONCE per block, then cache the fragments in
+  // registers (NCH <= 6 by eligibility) ----
   for (int idx = tid; idx < NCH * 512; idx += 256) {
     const int c = idx >> 9;
     const int r = (idx >> 5) & 15;
@@ -127,52 +127,79 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
     }
     lds_w[(c * 16 + r) * XLDW + kk] = v;
   }
-  // ---- stage x tile (raw, zero-padded; coalesced) ----
-  const int gt0 = t0 * stride - pad;
-  for (int idx = tid; idx < Cpg * SP; idx += 256) {
-    const int cig = idx / SP;
-    const int q = idx - cig * SP;
-    const unsigned short* xr = x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
-    const int gq = gt0 + q;
-    unsigned short v = (gq >= 0 && gq < Tin) ? xr[gq] : 0;
-    if (act_before) v = leaky_bf16x(v, slope);
-    lds_x[idx] = v;
-  }
   __syncthreads();
-
-  f32x4x acc[4] = {};
-  const int colb = wid * 64;
-  for (int c = 0; c < NCH; ++c) {
-    bf16x8x a0 = *(const bf16x8x*)(lds_w + (c * 16 + row) * XLDW + kk0);
+  bf16x8x af[6];
+  int xoff[6];  // this lane's B base offset (cig*SP + kt0) per chunk
+#pragma unroll
+  for (int c = 0; c < 6; ++c) {
+    if (c >= NCH) break;
+    af[c] = *(const bf16x8x*)(lds_w + (c * 16 + row) * XLDW + kk0);
     const int sub = c * 4 + (lane >> 4);
     const int cig = sub < NSUB ? sub / KT8 : 0;
     const int kt0 = sub < NSUB ? (sub - cig * KT8) * 8 : 0;
-    const unsigned short* xb = lds_x + cig * SP + kt0;
-#pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
-      const unsigned short* p = xb + (colb + nf * 16 + row) * stride;
-      bf16x8x bfr;
-      ((uint2*)&bfr)[0] = *(const uint2*)p;
-      ((uint2*)&bfr)[1] = *(const uint2*)(p + 4);
-      acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfr, acc[nf],
-                                                        0, 0, 0);
-    }
+    xoff[c] = cig * SP + kt0;
   }
 
+  // ---- t-tile loop: x re-staged per tile, weights stay in registers --
+  const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
+  const int tiles = (To + 255) >> 8;
+  for (int tt = blockIdx.x; tt < tiles; tt += gridDim.x) {
+    const int t0 = tt << 8;
+    // vectorized staging: interior 8-runs load+store 16 B at once
+    const int gt0 = t0 * stride - pad;
+    __syncthreads();  // previous tile's reads complete before overwrite
+    for (int idx = tid * 8; idx < Cpg * SP; idx += 2048) {
+      const int cig = idx / SP;
+      const int q = idx - cig * SP;
+      const unsigned short* xr =
+          x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+      const int gq = gt0 + q;
+      if (q + 8 <= SP && gq >= 0 && gq + 8 <= Tin && !act_before) {
+        *(uint4*)(lds_x + idx) = *(const uint4*)(xr + gq);
+      } else {
 #pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-    const int t = t0 + colb + nf * 16 + col;
-    if (t >= To) continue;
+        for (int j = 0; j < 8; ++j) {
+          if (q + j >= SP || cig != (idx + j) / SP) break;
+          unsigned short v =
+              (gq + j >= 0 && gq + j < Tin) ? xr[gq + j] : 0;
+          if (act_before) v = leaky_bf16x(v, slope);
+          lds_x[idx + j] = v;
+        }
+      }
+    }
+    __syncthreads();
+
+    f32x4x acc[4] = {};
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int co = co_base + row4 + r;
-      if (co >= co_end) continue;
-      float o = acc[nf][r] + (bias ? bias[co] : 0.f);
-      if (!act_before) o = apply_act(o, act, slope);
-      union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-      y[((int64_t)b * Cout + co) * To + t] = cv.s;
+    for (int c = 0; c < 6; ++c) {
+      if (c >= NCH) break;
+      const unsigned short* xb = lds_x + xoff[c];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const unsigned short* p = xb + (colb + nf * 16 + row) * stride;
+        bf16x8x bfr;
+        ((uint2*)&bfr)[0] = *(const uint2*)p;
+        ((uint2*)&bfr)[1] = *(const uint2*)(p + 4);
+        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[c], bfr,
+                                                          acc[nf], 0, 0, 0);
+      }
+    }
+
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int t = t0 + colb + nf * 16 + col;
+      if (t >= To) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co_base + row4 + r;
+        if (co >= co_end) continue;
+        float o = acc[nf][r] + (bias ? bias[co] : 0.f);
+        if (!act_before) o = apply_act(o, act, slope);
+        union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
+        y[((int64_t)b * Cout + co) * To + t] = cv.s;
+      }
     }
   }
 }
@@ -192,7 +219,7 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
                        unsigned short* __restrict__ gx, int B, int Cin,
                        int Tin, int Cout, int To, int K, int stride,
                        int pad, int groups, int nk, int NCH, int V,
-                       int coplog) {
+                       int coplog, int Umax) {
   constexpr int LDR = 20;  // [t][co] row stride (<=16 co + pad)
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   unsigned short* lds_w = smem;                // [NCH][16][XLDW]
@@ -202,7 +229,6 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int CoP = 1 << coplog;       // Copg (power of 2, 4..16)
   const int ipc = 32 >> coplog;      // i-values per 32-chunk
   const int g = blockIdx.y;
-  const int u0 = blockIdx.x * 256;
   const int b = blockIdx.z;
 
   const int tid = threadIdx.x;
@@ -211,7 +237,7 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int row = lane & 15;
   const int kk0 = (lane >> 4) * 8;
 
-  // ---- stage A (weights): rows (cig,phi), cols (i,co) co-fastest ----
+  // ---- stage A (weights) once: rows (cig,phi), cols (i,co) ----
   for (int idx = tid; idx < NCH * 512; idx += 256) {
     const int c = idx >> 9;
     const int r = (idx >> 5) & 15;
@@ -226,54 +252,80 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
       v = w[((int64_t)(g * CoP + co) * Cpg + cig) * K + k];
     lds_w[(c * 16 + r) * XLDW + kk] = v;
   }
-  // ---- stage gz tile transposed [t][co] (coalesced reads per co) ----
-  const int tlo = u0 - nk + 1;
-  for (int idx = tid; idx < CoP * V; idx += 256) {
-    const int co = idx / V;
-    const int v = idx - co * V;
-    const int t = tlo + v;
-    unsigned short e = 0;
-    if (t >= 0 && t < To)
-      e = gz[((int64_t)b * Cout + g * CoP + co) * To + t];
-    lds_gz[v * LDR + co] = e;
-  }
   __syncthreads();
-
-  f32x4x acc[4] = {};
-  const int colb = wid * 64;
-  for (int c = 0; c < NCH; ++c) {
-    bf16x8x a0 = *(const bf16x8x*)(lds_w + (c * 16 + row) * XLDW + kk0);
-    // fragment halves may straddle an i boundary when CoP == 4;
-    // i >= nk columns have zero A — clamp so the LDS read stays in range
+  bf16x8x af[6];
+  int off0[6], off1[6];  // B read offsets rel. to (vt*LDR): halves may
+#pragma unroll              // straddle an i boundary when CoP == 4
+  for (int c = 0; c < 6; ++c) {
+    if (c >= NCH) break;
+    af[c] = *(const bf16x8x*)(lds_w + (c * 16 + row) * XLDW + kk0);
+    // i >= nk columns have zero A — clamp keeps the LDS read in range
     const int i0 = min(c * ipc + (kk0 >> coplog), nk - 1);
-    const int co0 = kk0 & (CoP - 1);
     const int i1 = min(c * ipc + ((kk0 + 4) >> coplog), nk - 1);
-    const int co1 = (kk0 + 4) & (CoP - 1);
-#pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
-      const int vt = colb + nf * 16 + row + nk - 1;
-      bf16x8x bfr;
-      ((uint2*)&bfr)[0] = *(const uint2*)(lds_gz + (vt - i0) * LDR + co0);
-      ((uint2*)&bfr)[1] = *(const uint2*)(lds_gz + (vt - i1) * LDR + co1);
-      acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfr, acc[nf],
-                                                        0, 0, 0);
-    }
+    off0[c] = -i0 * LDR + (kk0 & (CoP - 1));
+    off1[c] = -i1 * LDR + ((kk0 + 4) & (CoP - 1));
   }
 
+  const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
+  const int utiles = (Umax + 255) >> 8;
+  for (int ut = blockIdx.x; ut < utiles; ut += gridDim.x) {
+    const int u0 = ut << 8;
+    const int tlo = u0 - nk + 1;
+    __syncthreads();
+    // vectorized gz staging: 8-wide coalesced loads, scattered writes
+    for (int idx = tid * 8; idx < CoP * V; idx += 2048) {
+      const int co = idx / V;
+      const int v = idx - co * V;
+      const unsigned short* zr =
+          gz + ((int64_t)b * Cout + g * CoP + co) * To;
+      const int t = tlo + v;
+      if (t >= 0 && t + 8 <= To && v + 8 <= V) {
+        const uint4 e = *(const uint4*)(zr + t);
+        const unsigned short* es = (const unsigned short*)&e;
 #pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-    const int u = u0 + colb + nf * 16 + col;
+        for (int j = 0; j < 8; ++j) lds_gz[(v + j) * LDR + co] = es[j];
+      } else {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int rg = row4 + r;
-      const int cig = rg / stride;
-      const int phi = rg - cig * stride;
-      const int j = u * stride + phi - pad;
-      if (j < 0 || j >= Tin) continue;
-      union { __bf16 bb; unsigned short s; } cv{(__bf16)acc[nf][r]};
-      gx[((int64_t)b * Cin + g * Cpg + cig) * Tin + j] = cv.s;
+        for (int j = 0; j < 8; ++j) {
+          if (v + j >= V) break;
+          const int tj = t + j;
+          lds_gz[(v + j) * LDR + co] =
+              (tj >= 0 && tj < To) ? zr[tj] : (unsigned short)0;
+        }
+      }
+    }
+    __syncthreads();
+
+    f32x4x acc[4] = {};
+#pragma unroll
+    for (int c = 0; c < 6; ++c) {
+      if (c >= NCH) break;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int vt = (colb + nf * 16 + row + nk - 1) * LDR;
+        bf16x8x bfr;
+        ((uint2*)&bfr)[0] = *(const uint2*)(lds_gz + vt + off0[c]);
+        ((uint2*)&bfr)[1] = *(const uint2*)(lds_gz + vt + off1[c]);
+        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[c], bfr,
+                                                          acc[nf], 0, 0, 0);
+      }
+    }
+
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int u = u0 + colb + nf * 16 + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int rg = row4 + r;
+        const int cig = rg / stride;
+        const int phi = rg - cig * stride;
+        const int j = u * stride + phi - pad;
+        if (j < 0 || j >= Tin) continue;
+        union { __bf16 bb; unsigned short s; } cv{(__bf16)acc[nf][r]};
+        gx[((int64_t)b * Cin + g * Cpg + cig) * Tin + j] = cv.s;
+      }
     }
   }
 }
@@ -324,56 +376,90 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
 
   f32x4x acc[3] = {};
 
+  // per-wave i block: i = wid*3 + nf; all three fragment windows fit in
+  // ONE 24-element (3x b128) read per (sc): window base i0 = wid*3
+  const int i0 = wid * NFRAG;
+  const int d0u = __builtin_amdgcn_readfirstlane(i0 & 7);
+  const int rowx = lane & 15;  // xp row = (cig*s + phi)
+
   for (int64_t cc = blockIdx.z; cc < total; cc += nsplit) {
     const int b = (int)(cc / tchunks);
     const int t0c = (int)(cc - (int64_t)b * tchunks) * TC;
-    // ---- stage gz [co][t] ----
-    for (int idx = tid; idx < 16 * TCP; idx += 256) {
+    __syncthreads();
+    // ---- stage gz [co][t]: 8-wide loads AND 16-B LDS writes ----
+    for (int idx = tid * 8; idx < 16 * TCP; idx += 2048) {
       const int co = idx / TCP;
       const int tt = idx - co * TCP;
-      unsigned short v = 0;
-      if (tt < TC && t0c + tt < To && co_base + co < co_end)
-        v = gz[((int64_t)b * Cout + co_base + co) * To + t0c + tt];
-      lds_gz[co * TCP + tt] = v;
+      union { unsigned short e[8]; uint4 q; } v = {};
+      if (co_base + co < co_end) {
+        const unsigned short* zr =
+            gz + ((int64_t)b * Cout + co_base + co) * To;
+        if (tt + 8 <= TC && t0c + tt + 8 <= To) {
+          v.q = *(const uint4*)(zr + t0c + tt);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (tt + j < TC && t0c + tt + j < To)
+              v.e[j] = zr[t0c + tt + j];
+        }
+      }
+      *(uint4*)(lds_gz + idx) = v.q;
     }
-    // ---- stage x phase-split (coalesced global, scattered LDS) ----
-    for (int idx = tid; idx < Cpg * SPW; idx += 256) {
+    // ---- stage x phase-split: 8-wide loads, paired 4-B LDS writes ----
+    for (int idx = tid * 8; idx < Cpg * SPW; idx += 2048) {
       const int cig = idx / SPW;
       const int qq = idx - cig * SPW;
-      const int gpos = t0c * stride - pad + qq;
-      const int phi = (gpos + pad) % stride;
-      const int m = (gpos + pad) / stride - t0c;
-      if (m < 0 || m >= TM) continue;
-      unsigned short v = 0;
-      if (gpos >= 0 && gpos < Tin) {
-        v = x[((int64_t)b * Cin + g * Cpg + cig) * Tin + gpos];
-        if (act_b) v = leaky_bf16x(v, slope);
+      const int gpos = t0c * stride - pad + qq;  // qq mult of 8
+      union { unsigned short e[8]; uint4 q; } v = {};
+      if (gpos >= 0 && gpos + 8 <= Tin) {
+        v.q = *(const uint4*)(x + ((int64_t)b * Cin + g * Cpg + cig) *
+                                      Tin + gpos);
+      } else {
+        const unsigned short* xr =
+            x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gpos + j >= 0 && gpos + j < Tin) v.e[j] = xr[gpos + j];
       }
-      lds_xp[(cig * stride + phi) * TM + m] = v;
+      if (act_b) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v.e[j] = leaky_bf16x(v.e[j], slope);
+      }
+      const int m0 = qq / stride;  // even (qq mult 8, stride 4 or 8)
+      if (stride == 4) {
+        // 8 gpos = phases 0..3 twice -> one u32 (m0, m0+1) per phase
+#pragma unroll
+        for (int phi = 0; phi < 4; ++phi) {
+          union { unsigned short e[2]; unsigned u; } pr;
+          pr.e[0] = v.e[phi];
+          pr.e[1] = v.e[phi + 4];
+          *(unsigned*)(lds_xp + (cig * 4 + phi) * TM + m0) = pr.u;
+        }
+      } else {  // stride 8: one element per phase
+#pragma unroll
+        for (int phi = 0; phi < 8; ++phi)
+          lds_xp[(cig * 8 + phi) * TM + m0] = v.e[phi];
+      }
     }
     __syncthreads();
 
-    // columns ordered n = i*16 + (cig*s + phi): each fragment covers
-    // all 16 (cig,phi) rows of xp at ONE i, so the window shift d = i&7
-    // is wave-uniform and extract8's switch stays scalar.
-    const int rowx = lane & 15;  // xp row = (cig*s + phi)
 #pragma unroll
     for (int sc = 0; sc < TC / 32; ++sc) {
       bf16x8x a0 =
           *(const bf16x8x*)(lds_gz + row * TCP + sc * 32 + kk0);
+      const unsigned short* p =
+          lds_xp + rowx * TM + (sc * 32 + kk0 + i0 - (i0 & 7));
+      unsigned wv[12];
+      *(uint4*)(wv + 0) = *(const uint4*)p;
+      *(uint4*)(wv + 4) = *(const uint4*)(p + 8);
+      *(uint4*)(wv + 8) = *(const uint4*)(p + 16);
 #pragma unroll
       for (int nf = 0; nf < NFRAG; ++nf) {
-        const int i = wid * NFRAG + nf;
-        const int du = __builtin_amdgcn_readfirstlane(i & 7);
-        const int start = sc * 32 + kk0 + i;
-        const unsigned short* p =
-            lds_xp + rowx * TM + (start - (i & 7));
-        bf16x8x bfr = extract8(p, du);
+        bf16x8x bfr = extract8w(wv, d0u + nf);
         acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfr, acc[nf],
                                                           0, 0, 0);
       }
     }
-    __syncthreads();
   }
 
   // epilogue: plain stores into this z-slice's private slab (summed on
@@ -423,9 +509,11 @@ bool xstage_fwd_eligible(int64_t Cin, int64_t Cout, int64_t K,
   if (groups <= 1 || dil != 1 || reflect) return false;
   const int Cpg = (int)(Cin / groups);
   const int Copg = (int)(Cout / groups);
-  // s*2B multiple of 8 keeps the b64 fragment reads aligned
+  // s*2B multiple of 8 keeps the b64 fragment reads aligned; the A
+  // fragments are register-cached (<= 6 chunks)
+  const int NCH = (Cpg * ((int)(K + 7) / 8) + 3) / 4;
   return (stride == 4 || stride == 8) && K <= 48 && K >= 9 && Cpg <= 8 &&
-         Copg <= 16;
+         Copg <= 16 && NCH <= 6;
 }
 
 torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
@@ -440,10 +528,15 @@ torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
   const int NSUB = Cpg * KT8;
   const int NCH = (NSUB + 3) / 4;
   const int SP = ((255 * (int)stride + K) + 7) / 8 * 8 + 8;
-  const size_t lds = (size_t)(NCH * 16 * XLDW + Cpg * SP) * 2;
+  const size_t lds = (size_t)(6 * 16 * XLDW + Cpg * SP) * 2;
   auto y = torch::empty({B, Cout, (int)To}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(xcdiv(To, 256), (int)groups * xcdiv(Copg, 16), B);
+  const int tiles = xcdiv(To, 256);
+  const int gyb = (int)groups * xcdiv(Copg, 16);
+  // few blocks per (y,z) loop many t-tiles: weights stay in registers
+  const int gx = (int)std::min<int64_t>(
+      tiles, std::max<int64_t>(1, 1024 / std::max(1, gyb * B)));
+  dim3 grid(gx, gyb, B);
   hipLaunchKernelGGL(xstage_fwd_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)w.data_ptr(), bias_p,
@@ -481,13 +574,16 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
   auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
   auto stream = at::hip::getCurrentHIPStream();
   const int Umax = ((int)Tin - 1 + (int)pad) / (int)stride + 1;
-  dim3 grid(xcdiv(Umax, 256), (int)groups, B);
+  const int utiles = xcdiv(Umax, 256);
+  const int gxd = (int)std::min<int64_t>(
+      utiles, std::max<int64_t>(1, 1024 / std::max(1, (int)groups * B)));
+  dim3 grid(gxd, (int)groups, B);
   hipLaunchKernelGGL(xstage_bwd_data_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
                      (unsigned short*)gx.data_ptr(), B, Cin, (int)Tin,
                      Cout, To, K, (int)stride, (int)pad, (int)groups, nk,
-                     NCH, V, coplog);
+                     NCH, V, coplog, Umax);
   return gx;
 }
 
