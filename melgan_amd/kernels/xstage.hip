@@ -102,7 +102,6 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   const int mt = blockIdx.y - g * mtiles;
   const int co_base = g * Copg + mt * 16;
   const int co_end = (g + 1) * Copg;
-  const int b = blockIdx.z;
   const int NSUB = Cpg * KT8;
 
   const int tid = threadIdx.x;
@@ -140,12 +139,16 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
     xoff[c] = cig * SP + kt0;
   }
 
-  // ---- t-tile loop: x re-staged per tile, weights stay in registers --
+  // ---- (b, t-tile) loop: x re-staged per tile, weights stay in
+  // registers across the whole walk ----
   const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
   const int tiles = (To + 255) >> 8;
-  for (int tt = blockIdx.x; tt < tiles; tt += gridDim.x) {
+  const int64_t total = (int64_t)B * tiles;
+  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
+    const int b = (int)(cc / tiles);
+    const int tt = (int)(cc - (int64_t)b * tiles);
     const int t0 = tt << 8;
     // vectorized staging: interior 8-runs load+store 16 B at once
     const int gt0 = t0 * stride - pad;
@@ -229,7 +232,6 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int CoP = 1 << coplog;       // Copg (power of 2, 4..16)
   const int ipc = 32 >> coplog;      // i-values per 32-chunk
   const int g = blockIdx.y;
-  const int b = blockIdx.z;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -270,8 +272,10 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
   const int utiles = (Umax + 255) >> 8;
-  for (int ut = blockIdx.x; ut < utiles; ut += gridDim.x) {
-    const int u0 = ut << 8;
+  const int64_t total = (int64_t)B * utiles;
+  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
+    const int b = (int)(cc / utiles);
+    const int u0 = (int)(cc - (int64_t)b * utiles) << 8;
     const int tlo = u0 - nk + 1;
     __syncthreads();
     // vectorized gz staging: 8-wide coalesced loads, scattered writes
@@ -533,10 +537,10 @@ torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
   auto stream = at::hip::getCurrentHIPStream();
   const int tiles = xcdiv(To, 256);
   const int gyb = (int)groups * xcdiv(Copg, 16);
-  // few blocks per (y,z) loop many t-tiles: weights stay in registers
-  const int gx = (int)std::min<int64_t>(
-      tiles, std::max<int64_t>(1, 1024 / std::max(1, gyb * B)));
-  dim3 grid(gx, gyb, B);
+  // persistent-ish blocks walk (b, t-tile); weights stay in registers
+  const int nz = (int)std::min<int64_t>(
+      (int64_t)B * tiles, std::max<int64_t>(1, 2048 / std::max(1, gyb)));
+  dim3 grid(1, gyb, nz);
   hipLaunchKernelGGL(xstage_fwd_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)w.data_ptr(), bias_p,
@@ -575,9 +579,10 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
   auto stream = at::hip::getCurrentHIPStream();
   const int Umax = ((int)Tin - 1 + (int)pad) / (int)stride + 1;
   const int utiles = xcdiv(Umax, 256);
-  const int gxd = (int)std::min<int64_t>(
-      utiles, std::max<int64_t>(1, 1024 / std::max(1, (int)groups * B)));
-  dim3 grid(gxd, (int)groups, B);
+  const int nz = (int)std::min<int64_t>(
+      (int64_t)B * utiles,
+      std::max<int64_t>(1, 2048 / std::max(1, (int)groups)));
+  dim3 grid(1, (int)groups, nz);
   hipLaunchKernelGGL(xstage_bwd_data_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
