@@ -539,7 +539,7 @@ torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
   const int gyb = (int)groups * xcdiv(Copg, 16);
   // persistent-ish blocks walk (b, t-tile); weights stay in registers
   const int nz = (int)std::min<int64_t>(
-      (int64_t)B * tiles, std::max<int64_t>(1, 2048 / std::max(1, gyb)));
+      (int64_t)B * tiles, std::max<int64_t>(1, 4096 / std::max(1, gyb)));
   dim3 grid(1, gyb, nz);
   hipLaunchKernelGGL(xstage_fwd_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)x.data_ptr(),
@@ -581,7 +581,7 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
   const int utiles = xcdiv(Umax, 256);
   const int nz = (int)std::min<int64_t>(
       (int64_t)B * utiles,
-      std::max<int64_t>(1, 2048 / std::max(1, (int)groups)));
+      std::max<int64_t>(1, 4096 / std::max(1, (int)groups)));
   dim3 grid(1, (int)groups, nz);
   hipLaunchKernelGGL(xstage_bwd_data_kernel, grid, dim3(256), lds, stream,
                      (const unsigned short*)gz.data_ptr(),
