@@ -25,6 +25,7 @@ SOURCES = [
     "stft.hip",
     "gemm.hip",
     "xstage.hip",
+    "pqmf.hip",
 ]
 
 

@@ -533,6 +533,13 @@ torch::Tensor multi_l1_fwd(torch::Tensor tbl, torch::Tensor chunks,
 void multi_l1_bwd(torch::Tensor tbl, torch::Tensor chunks, int64_t nt,
                   torch::Tensor gout, torch::ScalarType dt);
 
+// dedicated PQMF polyphase kernels (pqmf.hip)
+torch::Tensor pqmf_gather(torch::Tensor x, torch::Tensor fpad, int64_t KT,
+                          int64_t N, int64_t pad, double scale);
+torch::Tensor pqmf_scatter(torch::Tensor x, torch::Tensor fpad,
+                           int64_t KT, int64_t N, int64_t pad,
+                           double scale);
+
 // glds GEMM engine (gemm.hip)
 torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                               c10::optional<torch::Tensor> bias,
@@ -969,6 +976,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt", &gemm_bt);
   m.def("unpad_fold", &unpad_fold_py);
   m.def("conv1d_bwd_weight_gemm", &conv1d_bwd_weight_gemm);
+  m.def("pqmf_gather", &pqmf_gather);
+  m.def("pqmf_scatter", &pqmf_scatter);
   m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }

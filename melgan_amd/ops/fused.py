@@ -462,37 +462,88 @@ def fused_conv_transpose1d(mod, x: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
-# PQMF — fixed filters, no weight grads
+# PQMF — fixed filters, no weight grads.  Dedicated
+# polyphase kernels: `pqmf_gather` (stride-N filter-and-decimate) and
+# `pqmf_scatter` (polyphase upsample-and-filter) cover all four
+# directions with flipped-filter variants; the even taps count makes the
+# flipped pad equal the forward pad (KT-1-pad == pad).
 # ---------------------------------------------------------------------------
+def _pq_filters(mod, device):
+    """Cached fp32 zero-padded filter images for the fused kernels:
+    (H, flip(H), G, flip(G)) each (N, KTp)."""
+    cache = getattr(mod, "_pq_cache", None)
+    if cache is not None and cache[0] == device:
+        return cache[1]
+    N = mod.subbands
+    H = mod.analysis_filter.reshape(N, -1).float()
+    G = mod.synthesis_filter.reshape(N, -1).float()
+    KT = H.size(1)
+    KTp = (KT + 3) // 4 * 4 + 4
+    def pad_(f):
+        out = torch.zeros(N, KTp, device=device)
+        out[:, :KT] = f.to(device)
+        return out.contiguous()
+    filts = (pad_(H), pad_(H.flip(-1)), pad_(G), pad_(G.flip(-1)), KT)
+    mod._pq_cache = (device, filts)
+    return filts
+
+
 class PQMFAnalysisFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, filt, subbands, pad):
+    def forward(ctx, x, h, hflip, KT, subbands, pad):
         C = _ext.require_ext()
-        y = C.conv1d_fwd(x, filt, None, subbands, 1, pad, False, 1,
-                         ACT_NONE, 0.0, False)
-        ctx.save_for_backward(filt)
-        ctx.geom = (subbands, pad, x.size(2))
+        y = C.pqmf_gather(x, h, KT, subbands, pad, 1.0)
+        ctx.save_for_backward(hflip)
+        ctx.geom = (KT, subbands, pad)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         C = _ext.require_ext()
-        (filt,) = ctx.saved_tensors
-        subbands, pad, t_in = ctx.geom
-        gx = C.conv1d_bwd_data(gy.contiguous(), filt, subbands, 1, pad,
-                               False, 1, t_in)
-        return gx, None, None, None
+        (hflip,) = ctx.saved_tensors
+        KT, subbands, pad = ctx.geom
+        gx = C.pqmf_scatter(gy.contiguous(), hflip, KT, subbands,
+                            KT - 1 - pad, 1.0)
+        return gx, None, None, None, None, None
 
 
 def pqmf_analysis(mod, x: torch.Tensor) -> torch.Tensor:
+    C = _ext.ext()
+    x = x.contiguous()
+    if hasattr(C, "pqmf_gather") and x.dtype == torch.float32:
+        h, hflip, _, _, KT = _pq_filters(mod, x.device)
+        return PQMFAnalysisFn.apply(x, h, hflip, KT, mod.subbands, mod.pad)
     filt = mod.analysis_filter.to(x.dtype)
-    return PQMFAnalysisFn.apply(x.contiguous(), filt, mod.subbands, mod.pad)
+    y = _ext.require_ext().conv1d_fwd(x, filt, None, mod.subbands, 1,
+                                      mod.pad, False, 1, ACT_NONE, 0.0,
+                                      False)
+    return y
 
 
 class PQMFSynthesisFn(torch.autograd.Function):
     @staticmethod
+    def forward(ctx, x, g, gflip, KT, subbands, pad):
+        C = _ext.require_ext()
+        y = C.pqmf_scatter(x, g, KT, subbands, pad, float(subbands))
+        ctx.save_for_backward(gflip)
+        ctx.geom = (KT, subbands, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        C = _ext.require_ext()
+        (gflip,) = ctx.saved_tensors
+        KT, subbands, pad = ctx.geom
+        gx = C.pqmf_gather(gy.contiguous(), gflip, KT, subbands,
+                           KT - 1 - pad, float(subbands))
+        return gx, None, None, None, None, None
+
+
+class _PQMFSynthesisConvTFn(torch.autograd.Function):
+    """Fallback (pre-pqmf.hip extensions): synthesis on the convT path."""
+
+    @staticmethod
     def forward(ctx, x, filt_t, subbands, pad):
-        # filt_t: (N, 1, K) = flip(G)*N — conv_transpose weight layout
         C = _ext.require_ext()
         y = C.conv_transpose1d_fwd(x, filt_t, None, subbands, pad,
                                    subbands - 1, 0.0, False)
@@ -511,6 +562,12 @@ class PQMFSynthesisFn(torch.autograd.Function):
 
 
 def pqmf_synthesis(mod, x: torch.Tensor) -> torch.Tensor:
+    C = _ext.ext()
+    x = x.contiguous()
+    if hasattr(C, "pqmf_scatter") and x.dtype == torch.float32:
+        _, _, g, gflip, KT = _pq_filters(mod, x.device)
+        return PQMFSynthesisFn.apply(x, g, gflip, KT, mod.subbands,
+                                     mod.pad)
     key = "_fused_synth_w"
     w = getattr(mod, key, None)
     if w is None or w.dtype != x.dtype or w.device != x.device:
@@ -518,7 +575,7 @@ def pqmf_synthesis(mod, x: torch.Tensor) -> torch.Tensor:
         w = (mod.synthesis_filter * mod.subbands).flip(-1).to(x.dtype)
         setattr(mod, key, w)
     pad = mod.taps + 1 - 1 - mod.pad  # K-1-pad == pad for even taps
-    return PQMFSynthesisFn.apply(x.contiguous(), w, mod.subbands, pad)
+    return _PQMFSynthesisConvTFn.apply(x, w, mod.subbands, pad)
 
 
 # ---------------------------------------------------------------------------
