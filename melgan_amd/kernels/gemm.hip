@@ -158,39 +158,64 @@ gemm_bt_glds_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
 
 // im2col to the B^T image: out[(b*To + t)][ci*K + k] =
 //   pre(x[b][ci][t*stride + k*dil - pad]); zero for kk >= Cin*K (K-pad).
-// Vectorized: one 16-B store (8 kk) per thread; the (ci,k) split is
-// computed once and carried, not divided per element.
+// One 16-B store (8 kk) per item; the (ci,k) decode comes from an LDS
+// table built once per block, and the loop index splits with i32
+// divides only — the original per-element i64 div/mod chain ran the
+// kernel ~9x below HBM bandwidth.
 __global__ void im2col_bt_kernel(const unsigned short* __restrict__ x,
                                  unsigned short* __restrict__ out,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int Kp, bool act_before, float slope) {
+  extern __shared__ __attribute__((aligned(16))) unsigned tblsm[];
   const int K8 = Kp >> 3;
-  const int64_t total = (int64_t)Bb * To * K8;
-  const int64_t step = (int64_t)gridDim.x * blockDim.x;
   const int NK = Cin * K;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += step) {
-    const int k8 = (int)(i % K8);
-    const int64_t bt = i / K8;
-    const int t = (int)(bt % To);
-    const int b = (int)(bt / To);
-    int kk = k8 * 8;
-    int ci = kk / K;
-    int k = kk - ci * K;
-    const int tpos = t * stride - pad;
+  for (int k8 = threadIdx.x; k8 < K8; k8 += blockDim.x) {
+    const int kk = k8 * 8;
+    tblsm[k8] = kk < NK ? (((unsigned)(kk / K)) << 16) | (kk % K)
+                        : 0xffff0000u;
+  }
+  __syncthreads();
+  const int total = Bb * To * K8;  // chunks: < 2^31 for all our shapes
+  const int step = gridDim.x * blockDim.x;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += step) {
+    const int k8 = i % K8;
+    const int bt = i / K8;
+    const int b = bt / To;
+    const int t = bt - b * To;
+    const unsigned e0 = tblsm[k8];
+    int ci = (int)(e0 >> 16);
+    int k = (int)(e0 & 0xffffu);
     union { unsigned short e[8]; ushort4 q[2]; } v = {};
+    if (ci < Cin) {
+      const int tpos = t * stride - pad;
+      const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+      if (!reflect && !act_before && k + 8 <= K && dil == 1 &&
+          tpos + k >= 0 && tpos + k + 8 <= Tin) {
+        // fast path: 8 taps of one channel, fully interior
+        const unsigned short* p = xr + tpos + k;
+        v.e[0] = p[0]; v.e[1] = p[1]; v.e[2] = p[2]; v.e[3] = p[3];
+        v.e[4] = p[4]; v.e[5] = p[5]; v.e[6] = p[6]; v.e[7] = p[7];
+      } else {
+        int kk = ci * K + k;
 #pragma unroll
-    for (int j = 0; j < 8; ++j, ++kk, ++k) {
-      if (k >= K) { k = 0; ++ci; }
-      if (kk < NK) {
-        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-        unsigned short e = load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
-        if (act_before) e = leaky_bf16(e, slope);
-        v.e[j] = e;
+        for (int j = 0; j < 8; ++j, ++kk, ++k) {
+          if (k >= K) {
+            k = 0;
+            ++ci;
+            xr += Tin;
+          }
+          if (kk < NK) {
+            unsigned short e =
+                load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
+            if (act_before) e = leaky_bf16(e, slope);
+            v.e[j] = e;
+          }
+        }
       }
     }
-    ushort4* dst = (ushort4*)(out + bt * Kp + (int64_t)k8 * 8);
+    ushort4* dst = (ushort4*)(out + (int64_t)bt * Kp + k8 * 8);
     dst[0] = v.q[0];
     dst[1] = v.q[1];
   }
@@ -288,33 +313,50 @@ __global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int NKp, bool act_before, float slope) {
-  const int64_t NB = (int64_t)Bb * To;
-  const int64_t NB8 = NB >> 3;
-  const int64_t total = (int64_t)NKp * NB8;
-  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  extern __shared__ __attribute__((aligned(16))) unsigned tblsm[];
   const int NK = Cin * K;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += step) {
-    const int64_t bt8 = i % NB8;
-    const int kk = (int)(i / NB8);
+  for (int kk = threadIdx.x; kk < NKp; kk += blockDim.x)
+    tblsm[kk] = kk < NK ? (((unsigned)(kk / K)) << 16) | (kk % K)
+                        : 0xffff0000u;
+  __syncthreads();
+  const int NB = Bb * To;
+  const int NB8 = NB >> 3;
+  const int total = NKp * NB8;
+  const int step = gridDim.x * blockDim.x;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += step) {
+    const int bt8 = i % NB8;
+    const int kk = i / NB8;
     union { unsigned short e[8]; ushort4 q[2]; } v = {};
-    if (kk < NK) {
-      const int ci = kk / K;
-      const int k = kk - ci * K;
-      int b = (int)(bt8 * 8 / To);
-      int t = (int)(bt8 * 8 - (int64_t)b * To);
+    const unsigned e0 = tblsm[kk];
+    const int ci = (int)(e0 >> 16);
+    if (ci < Cin) {
+      const int k = (int)(e0 & 0xffffu);
+      int b = (bt8 * 8) / To;
+      int t = bt8 * 8 - b * To;
       const int koff = k * dil - pad;
+      const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
+      if (stride == 1 && !reflect && !act_before && t + 8 <= To &&
+          t + koff >= 0 && t + koff + 8 <= Tin) {
+        const unsigned short* p = xr + t + koff;
+        v.e[0] = p[0]; v.e[1] = p[1]; v.e[2] = p[2]; v.e[3] = p[3];
+        v.e[4] = p[4]; v.e[5] = p[5]; v.e[6] = p[6]; v.e[7] = p[7];
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-        unsigned short e = load_pad_bf16(xr, t * stride + koff, Tin,
-                                         reflect);
-        if (act_before) e = leaky_bf16(e, slope);
-        v.e[j] = e;
-        if (++t == To) { t = 0; ++b; }
+        for (int j = 0; j < 8; ++j) {
+          unsigned short e = load_pad_bf16(xr, t * stride + koff, Tin,
+                                           reflect);
+          if (act_before) e = leaky_bf16(e, slope);
+          v.e[j] = e;
+          if (++t == To) {
+            t = 0;
+            ++b;
+            xr += (int64_t)Cin * Tin;
+          }
+        }
       }
     }
-    ushort4* dst = (ushort4*)(out + kk * NB + bt8 * 8);
+    ushort4* dst = (ushort4*)(out + (int64_t)kk * NB + bt8 * 8);
     dst[0] = v.q[0];
     dst[1] = v.q[1];
   }
@@ -342,10 +384,11 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
   auto bt = torch::empty({(int64_t)Bb * To, (int64_t)Kp},
                          x.options());
   {
-    const int64_t n = (int64_t)Bb * To * Kp;
+    const int64_t n = (int64_t)Bb * To * Kp / 8;
+    const size_t tlds = (size_t)(Kp / 8) * 4;
     hipLaunchKernelGGL(im2col_bt_kernel,
-                       dim3((int)std::min<int64_t>(4096, (n + 255) / 256)),
-                       dim3(256), 0, stream,
+                       dim3((int)std::min<int64_t>(2048, (n + 255) / 256)),
+                       dim3(256), tlds, stream,
                        (const unsigned short*)x.data_ptr(),
                        (unsigned short*)bt.data_ptr(), Bb, Cin, Tin,
                        (int)To, (int)K, (int)stride, (int)dil, (int)pad,
@@ -403,10 +446,11 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   auto gzT = gz.permute({1, 0, 2}).reshape({Cout, Kred}).contiguous();
   auto btk = torch::empty({(int64_t)NKp, Kred}, x.options());
   {
-    const int64_t n = (int64_t)NKp * Kred;
+    const int64_t n = (int64_t)NKp * Kred / 8;
+    const size_t tlds = (size_t)NKp * 4;
     hipLaunchKernelGGL(im2col_tk_kernel,
-                       dim3((int)std::min<int64_t>(4096, (n + 255) / 256)),
-                       dim3(256), 0, stream,
+                       dim3((int)std::min<int64_t>(2048, (n + 255) / 256)),
+                       dim3(256), tlds, stream,
                        (const unsigned short*)x.data_ptr(),
                        (unsigned short*)btk.data_ptr(), Bb, Cin, Tin, To,
                        (int)K, (int)stride, (int)dil, (int)pad, reflect,
