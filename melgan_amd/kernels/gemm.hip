@@ -167,57 +167,50 @@ __global__ void im2col_bt_kernel(const unsigned short* __restrict__ x,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int Kp, bool act_before, float slope) {
+  // one ELEMENT per thread per iteration — consecutive lanes write (and
+  // mostly read) consecutive addresses, which is what the wave needs;
+  // the index split is divided once and CARRIED across grid-stride
+  // iterations, and the (ci,k) decode comes from an LDS table.
   extern __shared__ __attribute__((aligned(16))) unsigned tblsm[];
-  const int K8 = Kp >> 3;
   const int NK = Cin * K;
-  for (int k8 = threadIdx.x; k8 < K8; k8 += blockDim.x) {
-    const int kk = k8 * 8;
-    tblsm[k8] = kk < NK ? (((unsigned)(kk / K)) << 16) | (kk % K)
+  for (int kk = threadIdx.x; kk < Kp; kk += blockDim.x)
+    tblsm[kk] = kk < NK ? (((unsigned)(kk / K)) << 16) | (kk % K)
                         : 0xffff0000u;
-  }
   __syncthreads();
-  const int total = Bb * To * K8;  // chunks: < 2^31 for all our shapes
+  const int64_t total = (int64_t)Bb * To * Kp;
   const int step = gridDim.x * blockDim.x;
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += step) {
-    const int k8 = i % K8;
-    const int bt = i / K8;
-    const int b = bt / To;
-    const int t = bt - b * To;
-    const unsigned e0 = tblsm[k8];
-    int ci = (int)(e0 >> 16);
-    int k = (int)(e0 & 0xffffu);
-    union { unsigned short e[8]; ushort4 q[2]; } v = {};
+  const int stepK = step % Kp;
+  const int stepBt = step / Kp;
+  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int kk = (int)(i % Kp);
+  int bt = (int)(i / Kp);
+  int b = bt / To;
+  int t = bt - b * To;
+  for (; i < total; i += step) {
+    const unsigned e0 = tblsm[kk];
+    const int ci = (int)(e0 >> 16);
+    unsigned short v = 0;
     if (ci < Cin) {
-      const int tpos = t * stride - pad;
+      const int k = (int)(e0 & 0xffffu);
       const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-      if (!reflect && !act_before && k + 8 <= K && dil == 1 &&
-          tpos + k >= 0 && tpos + k + 8 <= Tin) {
-        // fast path: 8 taps of one channel, fully interior
-        const unsigned short* p = xr + tpos + k;
-        v.e[0] = p[0]; v.e[1] = p[1]; v.e[2] = p[2]; v.e[3] = p[3];
-        v.e[4] = p[4]; v.e[5] = p[5]; v.e[6] = p[6]; v.e[7] = p[7];
-      } else {
-        int kk = ci * K + k;
-#pragma unroll
-        for (int j = 0; j < 8; ++j, ++kk, ++k) {
-          if (k >= K) {
-            k = 0;
-            ++ci;
-            xr += Tin;
-          }
-          if (kk < NK) {
-            unsigned short e =
-                load_pad_bf16(xr, tpos + k * dil, Tin, reflect);
-            if (act_before) e = leaky_bf16(e, slope);
-            v.e[j] = e;
-          }
-        }
-      }
+      v = load_pad_bf16(xr, t * stride + k * dil - pad, Tin, reflect);
+      if (act_before) v = leaky_bf16(v, slope);
     }
-    ushort4* dst = (ushort4*)(out + (int64_t)bt * Kp + k8 * 8);
-    dst[0] = v.q[0];
-    dst[1] = v.q[1];
+    out[i] = v;
+    kk += stepK;
+    int btn = bt + stepBt;
+    if (kk >= Kp) {
+      kk -= Kp;
+      btn += 1;
+    }
+    t += btn - bt;
+    bt = btn;
+    if (t >= To) {
+      const int c = t / To;
+      b += c;
+      t -= c * To;
+    }
   }
 }
 
@@ -313,6 +306,7 @@ __global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
                                  int Bb, int Cin, int Tin, int To, int K,
                                  int stride, int dil, int pad, bool reflect,
                                  int NKp, bool act_before, float slope) {
+  // element-per-thread with carried indices (see im2col_bt_kernel)
   extern __shared__ __attribute__((aligned(16))) unsigned tblsm[];
   const int NK = Cin * K;
   for (int kk = threadIdx.x; kk < NKp; kk += blockDim.x)
@@ -320,45 +314,45 @@ __global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
                         : 0xffff0000u;
   __syncthreads();
   const int NB = Bb * To;
-  const int NB8 = NB >> 3;
-  const int total = NKp * NB8;
+  const int64_t total = (int64_t)NKp * NB;
   const int step = gridDim.x * blockDim.x;
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += step) {
-    const int bt8 = i % NB8;
-    const int kk = i / NB8;
-    union { unsigned short e[8]; ushort4 q[2]; } v = {};
-    const unsigned e0 = tblsm[kk];
+  const int stepBt = step % NB;
+  const int stepKk = step / NB;
+  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int bt = (int)(i % NB);
+  int kk = (int)(i / NB);
+  int b = bt / To;
+  int t = bt - b * To;
+  unsigned e0 = tblsm[kk];
+  for (; i < total; i += step) {
     const int ci = (int)(e0 >> 16);
+    unsigned short v = 0;
     if (ci < Cin) {
       const int k = (int)(e0 & 0xffffu);
-      int b = (bt8 * 8) / To;
-      int t = bt8 * 8 - b * To;
-      const int koff = k * dil - pad;
       const unsigned short* xr = x + ((int64_t)b * Cin + ci) * Tin;
-      if (stride == 1 && !reflect && !act_before && t + 8 <= To &&
-          t + koff >= 0 && t + koff + 8 <= Tin) {
-        const unsigned short* p = xr + t + koff;
-        v.e[0] = p[0]; v.e[1] = p[1]; v.e[2] = p[2]; v.e[3] = p[3];
-        v.e[4] = p[4]; v.e[5] = p[5]; v.e[6] = p[6]; v.e[7] = p[7];
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          unsigned short e = load_pad_bf16(xr, t * stride + koff, Tin,
-                                           reflect);
-          if (act_before) e = leaky_bf16(e, slope);
-          v.e[j] = e;
-          if (++t == To) {
-            t = 0;
-            ++b;
-            xr += (int64_t)Cin * Tin;
-          }
-        }
-      }
+      v = load_pad_bf16(xr, t * stride + k * dil - pad, Tin, reflect);
+      if (act_before) v = leaky_bf16(v, slope);
     }
-    ushort4* dst = (ushort4*)(out + (int64_t)kk * NB + bt8 * 8);
-    dst[0] = v.q[0];
-    dst[1] = v.q[1];
+    out[i] = v;
+    int btn = bt + stepBt;
+    kk += stepKk;
+    if (btn >= NB) {
+      btn -= NB;
+      kk += 1;
+    }
+    t += btn - bt;  // may be negative after wrap
+    bt = btn;
+    if (t >= To) {
+      const int c = t / To;
+      b += c;
+      t -= c * To;
+    } else if (t < 0) {
+      const int c = (-t + To - 1) / To;
+      b -= c;
+      t += c * To;
+    }
+    e0 = tblsm[kk < NKp ? kk : NKp - 1];
   }
 }
 
@@ -384,10 +378,10 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
   auto bt = torch::empty({(int64_t)Bb * To, (int64_t)Kp},
                          x.options());
   {
-    const int64_t n = (int64_t)Bb * To * Kp / 8;
-    const size_t tlds = (size_t)(Kp / 8) * 4;
+    const int64_t n = (int64_t)Bb * To * Kp;
+    const size_t tlds = (size_t)Kp * 4;
     hipLaunchKernelGGL(im2col_bt_kernel,
-                       dim3((int)std::min<int64_t>(2048, (n + 255) / 256)),
+                       dim3((int)std::min<int64_t>(4096, (n + 255) / 256)),
                        dim3(256), tlds, stream,
                        (const unsigned short*)x.data_ptr(),
                        (unsigned short*)bt.data_ptr(), Bb, Cin, Tin,
@@ -446,10 +440,10 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   auto gzT = gz.permute({1, 0, 2}).reshape({Cout, Kred}).contiguous();
   auto btk = torch::empty({(int64_t)NKp, Kred}, x.options());
   {
-    const int64_t n = (int64_t)NKp * Kred / 8;
+    const int64_t n = (int64_t)NKp * Kred;
     const size_t tlds = (size_t)NKp * 4;
     hipLaunchKernelGGL(im2col_tk_kernel,
-                       dim3((int)std::min<int64_t>(2048, (n + 255) / 256)),
+                       dim3((int)std::min<int64_t>(4096, (n + 255) / 256)),
                        dim3(256), tlds, stream,
                        (const unsigned short*)x.data_ptr(),
                        (unsigned short*)btk.data_ptr(), Bb, Cin, Tin, To,
