@@ -146,11 +146,7 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   const int row4 = (lane >> 4) * 4;
   const int tiles = (To + 255) >> 8;
   const int64_t total = (int64_t)B * tiles;
-  // contiguous chunk per block: consecutive iterations stay on one b's
-  // sequential tiles (L2-friendly staging)
-  const int64_t per = (total + gridDim.z - 1) / gridDim.z;
-  const int64_t cc1 = min((int64_t)(blockIdx.z + 1) * per, total);
-  for (int64_t cc = (int64_t)blockIdx.z * per; cc < cc1; ++cc) {
+  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
     const int b = (int)(cc / tiles);
     const int tt = (int)(cc - (int64_t)b * tiles);
     const int t0 = tt << 8;
@@ -277,9 +273,7 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int row4 = (lane >> 4) * 4;
   const int utiles = (Umax + 255) >> 8;
   const int64_t total = (int64_t)B * utiles;
-  const int64_t per = (total + gridDim.z - 1) / gridDim.z;
-  const int64_t cc1 = min((int64_t)(blockIdx.z + 1) * per, total);
-  for (int64_t cc = (int64_t)blockIdx.z * per; cc < cc1; ++cc) {
+  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
     const int b = (int)(cc / utiles);
     const int u0 = (int)(cc - (int64_t)b * utiles) << 8;
     const int tlo = u0 - nk + 1;
@@ -392,9 +386,7 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
   const int d0u = __builtin_amdgcn_readfirstlane(i0 & 7);
   const int rowx = lane & 15;  // xp row = (cig*s + phi)
 
-  const int64_t perz = (total + nsplit - 1) / nsplit;
-  const int64_t ccz1 = min((int64_t)(blockIdx.z + 1) * perz, total);
-  for (int64_t cc = (int64_t)blockIdx.z * perz; cc < ccz1; ++cc) {
+  for (int64_t cc = blockIdx.z; cc < total; cc += nsplit) {
     const int b = (int)(cc / tchunks);
     const int t0c = (int)(cc - (int64_t)b * tchunks) * TC;
     __syncthreads();
