@@ -100,6 +100,14 @@ class Trainer:
         self.timer = PhaseTimer(enabled=bool(hp.backend.get("phase_timing"))
                                 and self.device.type == "cuda")
         self.last_phases = {}
+        # side HIP stream: the multi-resolution STFT losses (rocFFT +
+        # reductions, small grids) overlap the discriminator convs; the
+        # backward nodes replay on their forward streams, so the overlap
+        # holds in backward too (SURVEY.md §5.1 / NOTES round-2 item)
+        use_ls = hp.backend.get("loss_stream")
+        self.loss_stream = torch.cuda.Stream() \
+            if (use_ls is None or use_ls) and self.device.type == "cuda" \
+            else None
 
     # ------------------------------------------------------------------
     def load(self, chkpt_path: str):
@@ -166,7 +174,23 @@ class Trainer:
         # ---- generator forward + losses --------------------------------
         tm.mark("g_fwd")
         self.dp_g.zero_grad(self.optim_g)
-        g_loss, parts, fake = self._forward_g(mel, audio)
+        if adversarial and self.loss_stream is not None:
+            # generator forward on the main stream; STFT losses on the
+            # side stream while the discriminator phase runs below
+            with torch.autocast("cuda", dtype=self.amp_dtype,
+                                enabled=self.use_amp):
+                fake_sub = self.generator(mel)
+            fake_sub = fake_sub.float()
+            fake = self.pqmf.synthesis(fake_sub) if self.multiband \
+                else fake_sub
+            self.loss_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self.loss_stream):
+                stft_loss, sub_loss = self._stft_losses(fake_sub, fake,
+                                                        audio)
+            g_loss = None  # joined after the discriminator forwards
+            parts = dict(stft=stft_loss, sub=sub_loss, adv=None, feat=None)
+        else:
+            g_loss, parts, fake = self._forward_g(mel, audio)
 
         # ---- discriminator forwards (both phases share D(real)) --------
         d_loss = None
@@ -205,6 +229,11 @@ class Trainer:
             # feature-matching targets reuse the d-phase D(real) features
             # (detached inside the loss) — D runs on real audio ONCE/step
             feat = L.feature_matching_loss(d_fake, d_real)
+            if g_loss is None:  # join the side-stream STFT losses
+                torch.cuda.current_stream().wait_stream(self.loss_stream)
+                stft = parts["stft"]
+                stft.record_stream(torch.cuda.current_stream())
+                g_loss = stft
             g_loss = g_loss + self.lambda_adv * adv + self.lambda_feat * feat
             parts["adv"], parts["feat"] = adv, feat
 
