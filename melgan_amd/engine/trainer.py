@@ -15,6 +15,7 @@ bf16 autocast on the model forward; all loss math fp32.
 
 from __future__ import annotations
 
+import contextlib
 import itertools
 import time
 
@@ -108,6 +109,12 @@ class Trainer:
         self.loss_stream = torch.cuda.Stream() \
             if (use_ls is None or use_ls) and self.device.type == "cuda" \
             else None
+        # the d-phase forward runs on its own stream so its BACKWARD
+        # (replayed there by autograd) overlaps the generator backward
+        use_ds = hp.backend.get("d_stream")
+        self.d_stream = torch.cuda.Stream() \
+            if (use_ds is None or use_ds) and self.device.type == "cuda" \
+            else None
 
     # ------------------------------------------------------------------
     def load(self, chkpt_path: str):
@@ -195,19 +202,6 @@ class Trainer:
         # ---- discriminator forwards (both phases share D(real)) --------
         d_loss = None
         if adversarial:
-            tm.mark("d_fwd")
-            self.dp_d.zero_grad(self.optim_d)
-            B = audio.size(0)
-            with torch.autocast("cuda", dtype=self.amp_dtype,
-                                enabled=self.use_amp):
-                # one 2B-batch forward instead of two B-batch forwards:
-                # identical math, twice the per-kernel tile fill
-                d_both = self.discriminator(
-                    torch.cat([audio.float(), fake.detach()], dim=0))
-            d_real = [[f[:B] for f in feats] for feats in d_both]
-            d_fake_det = [[f[B:] for f in feats] for feats in d_both]
-            d_loss = L.discriminator_loss(d_real, d_fake_det)
-
             # Freeze discriminator PARAMS for the generator-phase
             # adversarial forward: the g loss needs gradients THROUGH the
             # discriminator to `fake`, not INTO its weights.  Without
@@ -215,6 +209,8 @@ class Trainer:
             # kernels and (b) fires the disc's DP bucket hooks with
             # g-phase gradients, which would make the d-step's
             # all-reduce ship STALE gradients at world_size>1.
+            # It runs FIRST (on the main stream) so the weight-norm fold
+            # and GEMM operand caches fill before the side stream forks.
             tm.mark("g_adv_fwd")
             for p in self.discriminator.parameters():
                 p.requires_grad_(False)
@@ -225,6 +221,37 @@ class Trainer:
             finally:
                 for p in self.discriminator.parameters():
                     p.requires_grad_(True)
+
+            tm.mark("d_fwd")
+            self.dp_d.zero_grad(self.optim_d)
+            B = audio.size(0)
+            audio_f = audio.float()
+            fake_det = fake.detach()
+            ds = self.d_stream
+            main = torch.cuda.current_stream() if ds is not None else None
+            if ds is not None:
+                ds.wait_stream(main)
+            with torch.cuda.stream(ds) if ds is not None \
+                    else contextlib.nullcontext():
+                with torch.autocast("cuda", dtype=self.amp_dtype,
+                                    enabled=self.use_amp):
+                    # one 2B-batch forward instead of two B-batch
+                    # forwards: identical math, 2x per-kernel tile fill;
+                    # its backward replays on this stream, overlapping
+                    # the generator backward on the main stream
+                    d_both = self.discriminator(
+                        torch.cat([audio_f, fake_det], dim=0))
+            if ds is not None:
+                audio_f.record_stream(ds)
+                fake_det.record_stream(ds)
+                main.wait_stream(ds)
+                for feats in d_both:
+                    for f in feats:
+                        f.record_stream(main)
+            d_real = [[f[:B] for f in feats] for feats in d_both]
+            d_fake_det = [[f[B:] for f in feats] for feats in d_both]
+            d_loss = L.discriminator_loss(d_real, d_fake_det)
+
             adv = L.adversarial_g_loss(d_fake)
             # feature-matching targets reuse the d-phase D(real) features
             # (detached inside the loss) — D runs on real audio ONCE/step
@@ -248,10 +275,15 @@ class Trainer:
         # ---- discriminator update ---------------------------------------
         if adversarial:
             tm.mark("d_bwd")
-            d_loss.backward()
+            d_loss.backward()  # tower bwd kernels replay on d_stream,
+            # concurrent with the generator backward/step above
             tm.mark("d_comm_wait")
             self.dp_d.finish()
             tm.mark("d_opt")
+            if self.d_stream is not None:
+                # d grads were produced on d_stream; the fused Adam
+                # launches on the main stream
+                torch.cuda.current_stream().wait_stream(self.d_stream)
             self.optim_d.step()
         tm.mark(None)
         if tm.enabled:
