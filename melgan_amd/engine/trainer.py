@@ -144,6 +144,29 @@ class Trainer:
             self.discriminator, self.optim_g, self.optim_d, self.hp.hp_str)
 
     # ------------------------------------------------------------------
+    def _prewarm_disc(self):
+        """Materialise the discriminator's per-step weight folds and GEMM
+        operand caches on the CURRENT stream, so the d-stream fork and
+        the main-stream frozen forward can consume them concurrently
+        without a cross-stream race on first use."""
+        from melgan_amd.ops import _ext, fused
+
+        if _ext.ext() is None or self.device.type != "cuda":
+            return
+        dt = self.amp_dtype or torch.float32
+        for m in self.discriminator.modules():
+            if hasattr(m, "weight_v") and not getattr(m, "_removed", True):
+                w_c, _ = fused._wn_cached(m, dt)
+                if fused._gemm_fwd_eligible(
+                        w_c, m.in_channels, m.out_channels, m.kernel_size,
+                        m.stride, m.groups):
+                    fused._apad_cached(m, w_c)
+                if (m.stride == 1 and m.groups == 1 and
+                        fused._gemm_bwd_eligible(dt, m.in_channels,
+                                                 m.out_channels,
+                                                 m.kernel_size)):
+                    fused._apad_bwd_cached(m, w_c)
+
     def _stft_losses(self, fake_sub, fake, audio):
         """Multi-resolution STFT losses (full-band + sub-band mix)."""
         sc_full, mag_full = self.stft_full(fake.squeeze(1),
@@ -202,6 +225,30 @@ class Trainer:
         # ---- discriminator forwards (both phases share D(real)) --------
         d_loss = None
         if adversarial:
+            tm.mark("d_fwd")
+            self.dp_d.zero_grad(self.optim_d)
+            B = audio.size(0)
+            audio_f = audio.float()
+            fake_det = fake.detach()
+            ds = self.d_stream
+            main = torch.cuda.current_stream() if ds is not None else None
+            if ds is not None:
+                # weight folds / GEMM operand caches must exist BEFORE
+                # the fork — both streams' forwards consume them
+                self._prewarm_disc()
+                ds.wait_stream(main)
+            with torch.cuda.stream(ds) if ds is not None \
+                    else contextlib.nullcontext():
+                with torch.autocast("cuda", dtype=self.amp_dtype,
+                                    enabled=self.use_amp):
+                    # one 2B-batch forward instead of two B-batch
+                    # forwards: identical math, 2x per-kernel tile fill;
+                    # it runs on the d stream CONCURRENTLY with the
+                    # frozen g-phase forward below, and its backward
+                    # replays there, overlapping the generator backward
+                    d_both = self.discriminator(
+                        torch.cat([audio_f, fake_det], dim=0))
+
             # Freeze discriminator PARAMS for the generator-phase
             # adversarial forward: the g loss needs gradients THROUGH the
             # discriminator to `fake`, not INTO its weights.  Without
@@ -209,8 +256,6 @@ class Trainer:
             # kernels and (b) fires the disc's DP bucket hooks with
             # g-phase gradients, which would make the d-step's
             # all-reduce ship STALE gradients at world_size>1.
-            # It runs FIRST (on the main stream) so the weight-norm fold
-            # and GEMM operand caches fill before the side stream forks.
             tm.mark("g_adv_fwd")
             for p in self.discriminator.parameters():
                 p.requires_grad_(False)
@@ -222,25 +267,6 @@ class Trainer:
                 for p in self.discriminator.parameters():
                     p.requires_grad_(True)
 
-            tm.mark("d_fwd")
-            self.dp_d.zero_grad(self.optim_d)
-            B = audio.size(0)
-            audio_f = audio.float()
-            fake_det = fake.detach()
-            ds = self.d_stream
-            main = torch.cuda.current_stream() if ds is not None else None
-            if ds is not None:
-                ds.wait_stream(main)
-            with torch.cuda.stream(ds) if ds is not None \
-                    else contextlib.nullcontext():
-                with torch.autocast("cuda", dtype=self.amp_dtype,
-                                    enabled=self.use_amp):
-                    # one 2B-batch forward instead of two B-batch
-                    # forwards: identical math, 2x per-kernel tile fill;
-                    # its backward replays on this stream, overlapping
-                    # the generator backward on the main stream
-                    d_both = self.discriminator(
-                        torch.cat([audio_f, fake_det], dim=0))
             if ds is not None:
                 audio_f.record_stream(ds)
                 fake_det.record_stream(ds)
