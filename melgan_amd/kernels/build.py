@@ -26,6 +26,7 @@ SOURCES = [
     "gemm.hip",
     "xstage.hip",
     "pqmf.hip",
+    "gemm256.hip",
 ]
 
 

@@ -360,6 +360,13 @@ __global__ void im2col_tk_kernel(const unsigned short* __restrict__ x,
 
 using namespace melgan;
 
+
+// 256^2 two-wave-per-SIMD tier (gemm256.hip)
+bool gemm256_eligible(int64_t M, int64_t N, int64_t Kp);
+void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
+                 torch::Tensor y, int64_t M, int64_t N, int64_t Kp,
+                 int64_t To, int64_t act, double slope);
+
 static inline int gcdiv(int64_t a, int64_t b) {
   return (int)((a + b - 1) / b);
 }
@@ -396,6 +403,11 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
     bias_p = bias_f.data_ptr<float>();
   }
   const int64_t N = (int64_t)Bb * To;
+  if (gemm256_eligible(M, N, Kp)) {
+    gemm_bt_256(A_pad, bt, bias_p, y, M, N, Kp, To,
+                (int64_t)(act_before ? 0 : act), slope);
+    return y;
+  }
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);  // 32 KiB
   dim3 grid(gcdiv(N, GBN), gcdiv(M, GBM), 1);
   hipLaunchKernelGGL(gemm_bt_glds_kernel, grid, dim3(256), lds, stream,
@@ -414,6 +426,10 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt) {
   auto y = torch::empty({(int64_t)1, (int64_t)M, N},
                         A.options());
   auto stream = at::hip::getCurrentHIPStream();
+  if (gemm256_eligible(M, N, Kp)) {
+    gemm_bt_256(A, Bt, nullptr, y, M, N, Kp, N, 0, 0.0);
+    return y.view({M, N});
+  }
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);
   dim3 grid(gcdiv(N, GBN), gcdiv(M, GBM), 1);
   hipLaunchKernelGGL(gemm_bt_glds_kernel, grid, dim3(256), lds, stream,

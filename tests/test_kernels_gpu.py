@@ -821,7 +821,9 @@ def test_conv_shape_fuzz_bf16():
 
 
 @pytest.mark.parametrize("M,N,Kp", [
-    (128, 128, 32), (1024, 2048, 5120), (100, 300, 64), (64, 2048, 768)])
+    (128, 128, 32), (1024, 2048, 5120), (100, 300, 64), (64, 2048, 768),
+    # 256^2-tier shapes (M>=256, N>=256, Kp%64==0; incl. ragged tails)
+    (256, 512, 128), (512, 1000, 192), (300, 700, 1536), (2048, 512, 640)])
 def test_gemm_bt_engine(M, N, Kp):
     """glds GEMM engine vs torch matmul (asymmetric operands)."""
     import melgan_amd._C as C
