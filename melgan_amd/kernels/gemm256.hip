@@ -215,6 +215,15 @@ void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
   const int nwgx = q256cdiv(N, QBN);
   dim3 grid(nwgx, q256cdiv(M, QBM), 1);
   const size_t lds = 4 * QIMG * sizeof(unsigned short);  // 128 KiB
+  // dynamic LDS above the 64 KiB default needs an explicit opt-in
+  static const bool attr_ok = []() {
+    hipError_t e = hipFuncSetAttribute(
+        (const void*)gemm_bt_256_kernel,
+        hipFuncAttributeMaxDynamicSharedMemorySize,
+        4 * QIMG * sizeof(unsigned short));
+    return e == hipSuccess;
+  }();
+  TORCH_CHECK(attr_ok, "gemm256: hipFuncSetAttribute(LDS=128K) failed");
   hipLaunchKernelGGL(gemm_bt_256_kernel, grid, dim3(512), lds, stream,
                      (const unsigned short*)A.data_ptr(),
                      (const unsigned short*)Bt.data_ptr(), bias_p,
