@@ -75,7 +75,9 @@ gemm_bt_256_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
                    unsigned short* __restrict__ y,  // [Bb][M][To]
                    int M, int64_t N, int Kp, int To, int act, float slope,
                    int nwgx) {
-  extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
+  // 128 KiB STATIC LDS (dynamic shared memory is capped at 64 KiB on
+  // this runtime; static launches may use the full 160 KiB)
+  __shared__ __attribute__((aligned(16))) unsigned short smem[4 * QIMG];
 
   // T1: bijective XCD-aware remap of the N-tile index
   int bx = blockIdx.x;
@@ -214,17 +216,7 @@ void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
   auto stream = at::hip::getCurrentHIPStream();
   const int nwgx = q256cdiv(N, QBN);
   dim3 grid(nwgx, q256cdiv(M, QBM), 1);
-  const size_t lds = 4 * QIMG * sizeof(unsigned short);  // 128 KiB
-  // dynamic LDS above the 64 KiB default needs an explicit opt-in
-  static const bool attr_ok = []() {
-    hipError_t e = hipFuncSetAttribute(
-        (const void*)gemm_bt_256_kernel,
-        hipFuncAttributeMaxDynamicSharedMemorySize,
-        4 * QIMG * sizeof(unsigned short));
-    return e == hipSuccess;
-  }();
-  TORCH_CHECK(attr_ok, "gemm256: hipFuncSetAttribute(LDS=128K) failed");
-  hipLaunchKernelGGL(gemm_bt_256_kernel, grid, dim3(512), lds, stream,
+  hipLaunchKernelGGL(gemm_bt_256_kernel, grid, dim3(512), 0, stream,
                      (const unsigned short*)A.data_ptr(),
                      (const unsigned short*)Bt.data_ptr(), bias_p,
                      (unsigned short*)y.data_ptr(), (int)M, N, (int)Kp,
