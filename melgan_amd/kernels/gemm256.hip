@@ -28,8 +28,9 @@ typedef float f32x4q __attribute__((ext_vector_type(4)));
 
 constexpr int QBM = 256, QBN = 256, QBK = 64;
 constexpr int QROWB = QBK * 2;          // 128 B per image row
-constexpr int QIMG = QBM * QROWB;       // one operand image: 32 KiB
-// LDS: buf b at b*2*QIMG; A image first, B image second
+constexpr int QIMG = QBM * QBK;         // elements per operand image
+                                        // (256 x 64 bf16 = 32 KiB)
+// LDS: buf b at b*2*QIMG elements; A image first, B image second
 
 DEV_INLINE int qswz(int row, int byte_in_row) {
   return byte_in_row ^ ((row & 7) << 4);
