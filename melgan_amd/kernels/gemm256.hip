@@ -206,8 +206,13 @@ bool gemm256_enabled() {
 }
 
 bool gemm256_eligible(int64_t M, int64_t N, int64_t Kp) {
-  return gemm256_enabled() && M >= 256 && N >= 256 && Kp % QBK == 0 &&
-         Kp >= 128;
+  if (!(gemm256_enabled() && M >= 256 && N >= 256 && Kp % QBK == 0 &&
+        Kp >= 128))
+    return false;
+  // one 512-thread block per CU: the chip needs ~256 tiles to fill
+  // (under that the 128^2 engine's 4x block count wins)
+  const int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
+  return tiles >= 232;
 }
 
 // same contract as gemm.hip's engine call: y (Bb, M, To) pre-allocated
