@@ -212,9 +212,10 @@ bool gemm256_eligible(int64_t M, int64_t N, int64_t Kp) {
   // one 512-thread block per CU: the chip needs ~256 tiles to fill
   // (under that the 128^2 engine's 4x block count wins)
   const int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
-  // M-strip shapes (one M-tile) leave single blocks per CU exposed to
-  // the per-K-tile drain; the 128^2 engine's co-residency wins there
-  return tiles >= 232 && M >= 512;
+  // at 1 block/CU (128 KiB LDS) the per-K-tile drain is exposed unless
+  // multiple block-waves overlap; measured crossover ~2 waves of tiles,
+  // M-strip shapes also lose (single-M-tile -> poor L2 A reuse)
+  return tiles >= 448 && M >= 512;
 }
 
 // same contract as gemm.hip's engine call: y (Bb, M, To) pre-allocated
