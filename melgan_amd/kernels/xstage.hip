@@ -93,7 +93,7 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
                   int SP) {
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   unsigned short* lds_w = smem;                  // [NCH][16][XLDW]
-  unsigned short* lds_x = smem + 6 * 16 * XLDW;  // [Cpg][SP]
+  unsigned short* lds_x = smem + 6 * 16 * XLDW;  // 2 x [Cpg][SP] buffers
 
   const int Copg = Cout / groups;
   const int Cpg = Cin / groups;
@@ -139,46 +139,90 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
     xoff[c] = cig * SP + kt0;
   }
 
-  // ---- (b, t-tile) loop: x re-staged per tile, weights stay in
-  // registers across the whole walk ----
+  // ---- (b, t-tile) loop, double-buffered x: the NEXT tile's x is
+  // prefetched by async global_load_lds during the current tile's MFMA
+  // (raw s_barrier between tiles so the glds queue is not drained at
+  // every barrier); boundary tiles (zero padding — glds cannot mask)
+  // fall back to the synchronous vectorized staging path ----
   const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
   const int tiles = (To + 255) >> 8;
   const int64_t total = (int64_t)B * tiles;
-  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
-    const int b = (int)(cc / tiles);
-    const int tt = (int)(cc - (int64_t)b * tiles);
-    const int t0 = tt << 8;
-    // vectorized staging: interior 8-runs load+store 16 B at once
+  const int XBUF = ((Cpg * SP + 511) / 512) * 512;  // glds chunk pad
+  const int nch_g = XBUF / 512;  // 1-KiB glds chunks per tile
+
+  auto stage_sync = [&](unsigned short* dst, int b, int t0) {
     const int gt0 = t0 * stride - pad;
-    __syncthreads();  // previous tile's reads complete before overwrite
     for (int idx = tid * 8; idx < Cpg * SP; idx += 2048) {
       const int cig = idx / SP;
       const int q = idx - cig * SP;
       const unsigned short* xr =
           x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
       const int gq = gt0 + q;
-      if (q + 8 <= SP && gq >= 0 && gq + 8 <= Tin && !act_before) {
-        *(uint4*)(lds_x + idx) = *(const uint4*)(xr + gq);
+      if (gq >= 0 && gq + 8 <= Tin && !act_before) {
+        *(uint4*)(dst + idx) = *(const uint4*)(xr + gq);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          if (q + j >= SP || cig != (idx + j) / SP) break;
           unsigned short v =
               (gq + j >= 0 && gq + j < Tin) ? xr[gq + j] : 0;
           if (act_before) v = leaky_bf16x(v, slope);
-          lds_x[idx + j] = v;
+          dst[idx + j] = v;
         }
       }
     }
-    __syncthreads();
+  };
+  // async prefetch of a fully-interior tile (all source rows in range)
+  auto stage_glds = [&](unsigned short* dst, int b, int t0) {
+    const int gt0 = t0 * stride - pad;
+    const unsigned short* xb0 = x + ((int64_t)b * Cin + g * Cpg) * Tin;
+    for (int c = wid; c < nch_g; c += 4) {
+      const int base = c * 512;            // wave-uniform chunk base
+      int idx = base + lane * 8;
+      if (idx > Cpg * SP - 8) idx = Cpg * SP - 8;  // clamp tail lanes
+      const int cig = idx / SP;
+      const int q = idx - cig * SP;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)
+              (xb0 + (int64_t)cig * Tin + gt0 + q),
+          (__attribute__((address_space(3))) void*)(dst + base),
+          16, 0, 0);
+    }
+  };
+
+  int cur = 0;
+  bool have = false;  // buf[cur] staged for the first iteration?
+  for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
+    const int b = (int)(cc / tiles);
+    const int tt = (int)(cc - (int64_t)b * tiles);
+    const int t0 = tt << 8;
+    unsigned short* xcur = lds_x + cur * XBUF;
+    if (!have) {
+      __syncthreads();
+      stage_sync(xcur, b, t0);
+      __syncthreads();
+    }
+    // prefetch the next walk tile if it is interior
+    const int64_t cc2 = cc + gridDim.z;
+    bool pf = false;
+    int b2 = 0, t02 = 0;
+    if (cc2 < total) {
+      b2 = (int)(cc2 / tiles);
+      t02 = (int)(cc2 - (int64_t)b2 * tiles) << 8;
+      const int g2 = t02 * stride - pad;
+      // interior: every per-row read [g2, g2+SP) lies within [0, Tin)
+      if (g2 >= 0 && g2 + SP <= Tin) {
+        stage_glds(lds_x + (cur ^ 1) * XBUF, b2, t02);
+        pf = true;
+      }
+    }
 
     f32x4x acc[4] = {};
 #pragma unroll
     for (int c = 0; c < 6; ++c) {
       if (c >= NCH) break;
-      const unsigned short* xb = lds_x + xoff[c];
+      const unsigned short* xb = xcur + xoff[c];
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
         const unsigned short* p = xb + (colb + nf * 16 + row) * stride;
@@ -203,6 +247,16 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
         union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
         y[((int64_t)b * Cout + co) * To + t] = cv.s;
       }
+    }
+
+    if (pf) {
+      // drain this wave's prefetch, then raw barrier (no vmcnt attached)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      cur ^= 1;
+      have = true;
+    } else {
+      have = false;  // next iteration stages synchronously
     }
   }
 }
@@ -532,7 +586,8 @@ torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
   const int NSUB = Cpg * KT8;
   const int NCH = (NSUB + 3) / 4;
   const int SP = ((255 * (int)stride + K) + 7) / 8 * 8 + 8;
-  const size_t lds = (size_t)(6 * 16 * XLDW + Cpg * SP) * 2;
+  const int XBUF = ((Cpg * SP + 511) / 512) * 512;
+  const size_t lds = (size_t)(6 * 16 * XLDW + 2 * XBUF) * 2;
   auto y = torch::empty({B, Cout, (int)To}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
   const int tiles = xcdiv(To, 256);
