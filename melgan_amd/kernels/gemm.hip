@@ -81,7 +81,9 @@ gemm_bt_glds_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
                     const unsigned short* __restrict__ Bt,  // [N][Kp]
                     const float* __restrict__ bias,
                     unsigned short* __restrict__ y,  // [Bb][M][To]
-                    int M, int64_t N, int Kp, int To, int act, float slope) {
+                    int M, int64_t N, int Kp, int To, int act, float slope,
+                    int ilv) {  // >1: phase-interleaved convT store:
+                                // m=(phi,co) -> y[b][co][t*ilv+phi]
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   // [2 buffers][A tile | B tile], each tile 128*32 ushorts
   const int TILE = GBM * GBK;
@@ -151,7 +153,14 @@ gemm_bt_glds_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
         float o = acc[m][n][r] + (bias ? bias[gm] : 0.f);
         o = apply_act(o, act, slope);
         union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-        y[((int64_t)b * M + gm) * To + t] = cv.s;
+        if (ilv <= 1) {
+          y[((int64_t)b * M + gm) * To + t] = cv.s;
+        } else {
+          const int mo = M / ilv;       // Cout
+          const int phi = gm / mo;
+          const int co = gm - phi * mo;
+          y[((int64_t)b * mo + co) * To * ilv + t * ilv + phi] = cv.s;
+        }
       }
     }
 }
@@ -365,7 +374,7 @@ using namespace melgan;
 bool gemm256_eligible(int64_t M, int64_t N, int64_t Kp);
 void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
                  torch::Tensor y, int64_t M, int64_t N, int64_t Kp,
-                 int64_t To, int64_t act, double slope);
+                 int64_t To, int64_t act, double slope, int64_t ilv);
 
 static inline int gcdiv(int64_t a, int64_t b) {
   return (int)((a + b - 1) / b);
@@ -376,7 +385,8 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                               c10::optional<torch::Tensor> bias,
                               int64_t K, int64_t stride, int64_t dil,
                               int64_t pad, bool reflect, int64_t act,
-                              double slope, bool act_before, int64_t To) {
+                              double slope, bool act_before, int64_t To,
+                              int64_t ilv) {
   const int Bb = x.size(0), Cin = x.size(1), Tin = x.size(2);
   const int M = A_pad.size(0), Kp = A_pad.size(1);
   TORCH_CHECK(Kp % GBK == 0, "Kp must be a multiple of 32");
@@ -395,7 +405,9 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                        (int)To, (int)K, (int)stride, (int)dil, (int)pad,
                        reflect, Kp, act_before, (float)slope);
   }
-  auto y = torch::empty({Bb, M, (int)To}, x.options());
+  auto y = ilv > 1
+      ? torch::empty({Bb, M / ilv, (int64_t)To * ilv}, x.options())
+      : torch::empty({Bb, M, (int)To}, x.options());
   torch::Tensor bias_f;
   const float* bias_p = nullptr;
   if (bias.has_value()) {
@@ -405,7 +417,7 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
   const int64_t N = (int64_t)Bb * To;
   if (gemm256_eligible(M, N, Kp)) {
     gemm_bt_256(A_pad, bt, bias_p, y, M, N, Kp, To,
-                (int64_t)(act_before ? 0 : act), slope);
+                (int64_t)(act_before ? 0 : act), slope, ilv);
     return y;
   }
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);  // 32 KiB
@@ -414,7 +426,7 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                      (const unsigned short*)A_pad.data_ptr(),
                      (const unsigned short*)bt.data_ptr(), bias_p,
                      (unsigned short*)y.data_ptr(), M, N, Kp, (int)To,
-                     (int)(act_before ? 0 : act), (float)slope);
+                     (int)(act_before ? 0 : act), (float)slope, (int)ilv);
   return y;
 }
 
@@ -427,7 +439,7 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt) {
                         A.options());
   auto stream = at::hip::getCurrentHIPStream();
   if (gemm256_eligible(M, N, Kp)) {
-    gemm_bt_256(A, Bt, nullptr, y, M, N, Kp, N, 0, 0.0);
+    gemm_bt_256(A, Bt, nullptr, y, M, N, Kp, N, 0, 0.0, 0);
     return y.view({M, N});
   }
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);
@@ -436,7 +448,7 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt) {
                      (const unsigned short*)A.data_ptr(),
                      (const unsigned short*)Bt.data_ptr(), nullptr,
                      (unsigned short*)y.data_ptr(), M, N, Kp, (int)N, 0,
-                     0.f);
+                     0.f, 0);
   return y.view({M, N});
 }
 

@@ -75,7 +75,7 @@ gemm_bt_256_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
                    const float* __restrict__ bias,
                    unsigned short* __restrict__ y,  // [Bb][M][To]
                    int M, int64_t N, int Kp, int To, int act, float slope,
-                   int nwgx) {
+                   int nwgx, int ilv) {
   // 128 KiB STATIC LDS (dynamic shared memory is capped at 64 KiB on
   // this runtime; static launches may use the full 160 KiB)
   __shared__ __attribute__((aligned(16))) unsigned short smem[4 * QIMG];
@@ -184,7 +184,14 @@ gemm_bt_256_kernel(const unsigned short* __restrict__ A,   // [M][Kp]
         float o = acc[mf][nf][r] + (bias ? bias[gm] : 0.f);
         o = apply_act(o, act, slope);
         union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
-        y[((int64_t)b * M + gm) * To + t] = cv.s;
+        if (ilv <= 1) {
+          y[((int64_t)b * M + gm) * To + t] = cv.s;
+        } else {
+          const int mo = M / ilv;
+          const int phi = gm / mo;
+          const int co = gm - phi * mo;
+          y[((int64_t)b * mo + co) * To * ilv + t * ilv + phi] = cv.s;
+        }
       }
     }
 }
@@ -221,7 +228,7 @@ bool gemm256_eligible(int64_t M, int64_t N, int64_t Kp) {
 // same contract as gemm.hip's engine call: y (Bb, M, To) pre-allocated
 void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
                  torch::Tensor y, int64_t M, int64_t N, int64_t Kp,
-                 int64_t To, int64_t act, double slope) {
+                 int64_t To, int64_t act, double slope, int64_t ilv) {
   auto stream = at::hip::getCurrentHIPStream();
   const int nwgx = q256cdiv(N, QBN);
   dim3 grid(nwgx, q256cdiv(M, QBM), 1);
@@ -229,5 +236,5 @@ void gemm_bt_256(torch::Tensor A, torch::Tensor Bt, const float* bias_p,
                      (const unsigned short*)A.data_ptr(),
                      (const unsigned short*)Bt.data_ptr(), bias_p,
                      (unsigned short*)y.data_ptr(), (int)M, N, (int)Kp,
-                     (int)To, (int)act, (float)slope, nwgx);
+                     (int)To, (int)act, (float)slope, nwgx, (int)ilv);
 }

@@ -128,7 +128,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
                   ) // stride + 1
             y = C.conv1d_fwd_gemm(x, apad, b_c, w_c.size(2), stride,
                                   dilation, pad, reflect, act, slope,
-                                  act_before, To)
+                                  act_before, To, 0)
         else:
             y = C.conv1d_fwd(x, w_c, b_c, stride, dilation, pad, reflect,
                              groups, act, slope, act_before)
@@ -166,7 +166,7 @@ class FusedWNConv1dFn(torch.autograd.Function):
                 Tpad = x.size(2) + 2 * pad
                 gxpad = C.conv1d_fwd_gemm(
                     gz, ctx.apad_bwd, None, K, 1, dilation,
-                    (K - 1) * dilation, False, 0, 0.0, False, Tpad)
+                    (K - 1) * dilation, False, 0, 0.0, False, Tpad, 0)
                 gx = C.unpad_fold(gxpad, x.size(2), pad, reflect) \
                     if pad > 0 else gxpad
             else:
@@ -224,7 +224,7 @@ def fused_conv1d(mod, x: torch.Tensor) -> torch.Tensor:
             return C.conv1d_fwd_gemm(x, apad, b_c, mod.kernel_size,
                                      mod.stride, mod.dilation, mod.pad,
                                      mod.pad_mode == "reflect", mod.act,
-                                     mod.act_slope, mod.act_before, To)
+                                     mod.act_slope, mod.act_before, To, 0)
         return C.conv1d_fwd(x, w_c, b_c, mod.stride, mod.dilation, mod.pad,
                             mod.pad_mode == "reflect", mod.groups, mod.act,
                             mod.act_slope, mod.act_before)
@@ -333,11 +333,11 @@ def _biasph_cached(mod):
 
 
 def _convt_gemm_fwd(C, x, aph, bias_ph, cout, stride, slope, act_before):
-    B, _, T = x.shape
-    yph = C.conv1d_fwd_gemm(x, aph, bias_ph, 3, 1, 1, 1, False,
-                            ACT_NONE, slope, act_before, T)
-    return yph.view(B, stride, cout, T).permute(0, 2, 3, 1) \
-        .reshape(B, cout, T * stride)
+    # ilv=stride: the GEMM epilogue scatters m=(phi,co) straight to
+    # y[b][co][t*stride+phi] — no post-permute pass
+    return C.conv1d_fwd_gemm(x, aph, bias_ph, 3, 1, 1, 1, False,
+                             ACT_NONE, slope, act_before, x.size(2),
+                             stride)
 
 
 class ConvTGemmFn(torch.autograd.Function):
@@ -363,7 +363,7 @@ class ConvTGemmFn(torch.autograd.Function):
         gx = gv = gg = gb = None
         if ctx.needs_input_grad[0]:
             gx = C.conv1d_fwd_gemm(gy, a2, None, K, stride, 1, pad, False,
-                                   ACT_NONE, 0.0, False, T)
+                                   ACT_NONE, 0.0, False, T, 0)
             if act_before:
                 C.leaky_bwd_inplace(gx, x, slope)
         if ctx.needs_input_grad[1]:
