@@ -545,7 +545,8 @@ torch::Tensor conv1d_fwd_gemm(torch::Tensor x, torch::Tensor A_pad,
                               c10::optional<torch::Tensor> bias,
                               int64_t K, int64_t stride, int64_t dil,
                               int64_t pad, bool reflect, int64_t act,
-                              double slope, bool act_before, int64_t To);
+                              double slope, bool act_before, int64_t To,
+                              int64_t ilv);
 torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt);
 torch::Tensor unpad_fold_py(torch::Tensor gxpad, int64_t Tin, int64_t pad,
                             bool reflect);
