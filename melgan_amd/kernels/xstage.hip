@@ -127,8 +127,14 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
     lds_w[(c * 16 + r) * XLDW + kk] = v;
   }
   __syncthreads();
+  (void)SP;  // superseded by the XROW layout below
+  const bool bfold = To < 256 && (To & (To - 1)) == 0 && To >= 16;
+  const int TT = bfold ? To : 256;     // t per tile
+  const int NB = 256 / TT;             // batch items per tile
+  const int SPW = (TT - 1) * stride + KT8 * 8 + 8;  // per-(b,cig) span
+  const int XROW = (SPW + 7) / 8 * 8;
   bf16x8x af[6];
-  int xoff[6];  // this lane's B base offset (cig*SP + kt0) per chunk
+  int xoff[6];  // this lane's B base offset (cig*XROW + kt0) per chunk
 #pragma unroll
   for (int c = 0; c < 6; ++c) {
     if (c >= NCH) break;
@@ -136,56 +142,70 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
     const int sub = c * 4 + (lane >> 4);
     const int cig = sub < NSUB ? sub / KT8 : 0;
     const int kt0 = sub < NSUB ? (sub - cig * KT8) * 8 : 0;
-    xoff[c] = cig * SP + kt0;
+    xoff[c] = cig * XROW + kt0;
   }
 
-  // ---- (b, t-tile) loop, double-buffered x: the NEXT tile's x is
-  // prefetched by async global_load_lds during the current tile's MFMA
-  // (raw s_barrier between tiles so the glds queue is not drained at
-  // every barrier); boundary tiles (zero padding — glds cannot mask)
-  // fall back to the synchronous vectorized staging path ----
+  // ---- (b-group, t-tile) loop, double-buffered x: the NEXT tile's x
+  // is prefetched by async global_load_lds during the current tile's
+  // MFMA (raw s_barrier between tiles so the glds queue is not drained
+  // at every barrier); boundary tiles (zero padding — glds cannot mask)
+  // fall back to the synchronous vectorized staging path.
+  // Small-To layers (L3/L4 scales: To in {16,32,64,128}) fold NB =
+  // 256/To batch items into the column dimension, so the 256-col tile
+  // is full instead of (1 - To/256) masked ----
   const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
-  const int tiles = (To + 255) >> 8;
-  const int64_t total = (int64_t)B * tiles;
-  const int XBUF = ((Cpg * SP + 511) / 512) * 512;  // glds chunk pad
+  const int tlog = 31 - __clz(TT);
+  const int tiles = bfold ? 1 : (To + 255) >> 8;
+  const int64_t total = (int64_t)(B / NB + (B % NB ? 1 : 0)) * tiles;
+  const int XBUF = ((NB * Cpg * XROW + 511) / 512) * 512;
   const int nch_g = XBUF / 512;  // 1-KiB glds chunks per tile
 
-  auto stage_sync = [&](unsigned short* dst, int b, int t0) {
+  // LDS x layout: [b_off][cig][XROW]
+  auto stage_sync = [&](unsigned short* dst, int bg, int t0) {
     const int gt0 = t0 * stride - pad;
-    for (int idx = tid * 8; idx < Cpg * SP; idx += 2048) {
-      const int cig = idx / SP;
-      const int q = idx - cig * SP;
-      const unsigned short* xr =
-          x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
-      const int gq = gt0 + q;
-      if (gq >= 0 && gq + 8 <= Tin && !act_before) {
-        *(uint4*)(dst + idx) = *(const uint4*)(xr + gq);
-      } else {
+    for (int idx = tid * 8; idx < NB * Cpg * XROW; idx += 2048) {
+      const int bo = idx / (Cpg * XROW);
+      const int rem = idx - bo * Cpg * XROW;
+      const int cig = rem / XROW;
+      const int q = rem - cig * XROW;
+      const int b = bg * NB + bo;
+      unsigned short v8[8] = {};
+      if (b < B) {
+        const unsigned short* xr =
+            x + ((int64_t)b * Cin + g * Cpg + cig) * Tin;
+        const int gq = gt0 + q;
+        if (gq >= 0 && gq + 8 <= Tin && !act_before) {
+          *(uint4*)(dst + idx) = *(const uint4*)(xr + gq);
+          continue;
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           unsigned short v =
               (gq + j >= 0 && gq + j < Tin) ? xr[gq + j] : 0;
           if (act_before) v = leaky_bf16x(v, slope);
-          dst[idx + j] = v;
+          v8[j] = v;
         }
       }
+      *(uint4*)(dst + idx) = *(const uint4*)v8;
     }
   };
   // async prefetch of a fully-interior tile (all source rows in range)
-  auto stage_glds = [&](unsigned short* dst, int b, int t0) {
+  auto stage_glds = [&](unsigned short* dst, int bg, int t0) {
     const int gt0 = t0 * stride - pad;
-    const unsigned short* xb0 = x + ((int64_t)b * Cin + g * Cpg) * Tin;
     for (int c = wid; c < nch_g; c += 4) {
       const int base = c * 512;            // wave-uniform chunk base
       int idx = base + lane * 8;
-      if (idx > Cpg * SP - 8) idx = Cpg * SP - 8;  // clamp tail lanes
-      const int cig = idx / SP;
-      const int q = idx - cig * SP;
+      if (idx > NB * Cpg * XROW - 8) idx = NB * Cpg * XROW - 8;
+      const int bo = idx / (Cpg * XROW);
+      const int rem = idx - bo * Cpg * XROW;
+      const int cig = rem / XROW;
+      const int q = rem - cig * XROW;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)
-              (xb0 + (int64_t)cig * Tin + gt0 + q),
+              (x + ((int64_t)(bg * NB + bo) * Cin + g * Cpg + cig) * Tin +
+               gt0 + q),
           (__attribute__((address_space(3))) void*)(dst + base),
           16, 0, 0);
     }
@@ -194,26 +214,25 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   int cur = 0;
   bool have = false;  // buf[cur] staged for the first iteration?
   for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
-    const int b = (int)(cc / tiles);
-    const int tt = (int)(cc - (int64_t)b * tiles);
-    const int t0 = tt << 8;
+    const int bg = (int)(cc / tiles);
+    const int tt = (int)(cc - (int64_t)bg * tiles);
+    const int t0 = tt * TT;
     unsigned short* xcur = lds_x + cur * XBUF;
     if (!have) {
       __syncthreads();
-      stage_sync(xcur, b, t0);
+      stage_sync(xcur, bg, t0);
       __syncthreads();
     }
     // prefetch the next walk tile if it is interior
     const int64_t cc2 = cc + gridDim.z;
     bool pf = false;
-    int b2 = 0, t02 = 0;
     if (cc2 < total) {
-      b2 = (int)(cc2 / tiles);
-      t02 = (int)(cc2 - (int64_t)b2 * tiles) << 8;
+      const int bg2 = (int)(cc2 / tiles);
+      const int t02 = (int)(cc2 - (int64_t)bg2 * tiles) * TT;
       const int g2 = t02 * stride - pad;
-      // interior: every per-row read [g2, g2+SP) lies within [0, Tin)
-      if (g2 >= 0 && g2 + SP <= Tin) {
-        stage_glds(lds_x + (cur ^ 1) * XBUF, b2, t02);
+      // interior: every per-row read in range AND all NB batch rows real
+      if (g2 >= 0 && g2 + XROW <= Tin && (bg2 + 1) * NB <= B) {
+        stage_glds(lds_x + (cur ^ 1) * XBUF, bg2, t02);
         pf = true;
       }
     }
@@ -222,10 +241,13 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
     for (int c = 0; c < 6; ++c) {
       if (c >= NCH) break;
-      const unsigned short* xb = xcur + xoff[c];
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        const unsigned short* p = xb + (colb + nf * 16 + row) * stride;
+        const int cl = colb + nf * 16 + row;     // tile column
+        const int bo = cl >> tlog;               // 0 when !bfold
+        const int tl = cl & (TT - 1);
+        const unsigned short* p = xcur + (bo * Cpg) * XROW + xoff[c] +
+                                  tl * stride;
         bf16x8x bfr;
         ((uint2*)&bfr)[0] = *(const uint2*)p;
         ((uint2*)&bfr)[1] = *(const uint2*)(p + 4);
@@ -236,8 +258,10 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
 
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      const int t = t0 + colb + nf * 16 + col;
-      if (t >= To) continue;
+      const int cl = colb + nf * 16 + col;
+      const int b = bg * NB + (cl >> tlog);
+      const int t = t0 + (cl & (TT - 1));
+      if (t >= To || b >= B) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int co = co_base + row4 + r;
@@ -586,7 +610,13 @@ torch::Tensor xstage_fwd(torch::Tensor x, torch::Tensor w,
   const int NSUB = Cpg * KT8;
   const int NCH = (NSUB + 3) / 4;
   const int SP = ((255 * (int)stride + K) + 7) / 8 * 8 + 8;
-  const int XBUF = ((Cpg * SP + 511) / 512) * 512;
+  // mirror the kernel's tile geometry for the LDS size
+  const bool bfold = To < 256 && (To & (To - 1)) == 0 && To >= 16;
+  const int TT = bfold ? (int)To : 256;
+  const int NB = 256 / TT;
+  const int KT8h = (K + 7) / 8;
+  const int XROW = (((TT - 1) * (int)stride + KT8h * 8 + 8) + 7) / 8 * 8;
+  const int XBUF = ((NB * Cpg * XROW + 511) / 512) * 512;
   const size_t lds = (size_t)(6 * 16 * XLDW + 2 * XBUF) * 2;
   auto y = torch::empty({B, Cout, (int)To}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
