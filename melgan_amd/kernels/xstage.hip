@@ -349,32 +349,50 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
   const int colb = wid * 64;
   const int col = lane & 15;
   const int row4 = (lane >> 4) * 4;
-  const int utiles = (Umax + 255) >> 8;
-  const int64_t total = (int64_t)B * utiles;
+  // small-Umax layers: fold NB = 256/TT batch items into the columns
+  // (TT = smallest pow2 >= Umax, min 16) — mirrors xstage_fwd
+  int TT = 256;
+  while (TT >= 32 && (TT >> 1) >= Umax) TT >>= 1;
+  const int NB = 256 / TT;
+  const int tlog = 31 - __clz(TT);
+  const int utiles = (Umax + TT - 1) / TT;
+  const int VR = ((TT + nk + 7) / 8) * 8 + 8;  // per-b staged t rows
+  const int64_t total =
+      (int64_t)(B / NB + (B % NB ? 1 : 0)) * utiles;
   for (int64_t cc = blockIdx.z; cc < total; cc += gridDim.z) {
-    const int b = (int)(cc / utiles);
-    const int u0 = (int)(cc - (int64_t)b * utiles) << 8;
+    const int bg = (int)(cc / utiles);
+    const int u0 = (int)(cc - (int64_t)bg * utiles) * TT;
     const int tlo = u0 - nk + 1;
     __syncthreads();
-    // vectorized gz staging: 8-wide coalesced loads, scattered writes
-    for (int idx = tid * 8; idx < CoP * V; idx += 2048) {
-      const int co = idx / V;
-      const int v = idx - co * V;
+    // vectorized gz staging: 8-wide coalesced loads, scattered writes;
+    // LDS layout [b_off][v][LDR]
+    for (int idx = tid * 8; idx < NB * CoP * VR; idx += 2048) {
+      const int bo = idx / (CoP * VR);
+      const int rem = idx - bo * CoP * VR;
+      const int co = rem / VR;
+      const int v = rem - co * VR;
+      const int b = bg * NB + bo;
+      unsigned short* dst = lds_gz + (bo * VR + v) * LDR + co;
+      if (b >= B) {
+#pragma unroll
+        for (int j = 0; j < 8 && v + j < VR; ++j) dst[j * LDR] = 0;
+        continue;
+      }
       const unsigned short* zr =
           gz + ((int64_t)b * Cout + g * CoP + co) * To;
       const int t = tlo + v;
-      if (t >= 0 && t + 8 <= To && v + 8 <= V) {
+      if (t >= 0 && t + 8 <= To && v + 8 <= VR) {
         const uint4 e = *(const uint4*)(zr + t);
         const unsigned short* es = (const unsigned short*)&e;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) lds_gz[(v + j) * LDR + co] = es[j];
+        for (int j = 0; j < 8; ++j) dst[j * LDR] = es[j];
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          if (v + j >= V) break;
+          if (v + j >= VR) break;
           const int tj = t + j;
-          lds_gz[(v + j) * LDR + co] =
-              (tj >= 0 && tj < To) ? zr[tj] : (unsigned short)0;
+          dst[j * LDR] = (tj >= 0 && tj < To) ? zr[tj]
+                                              : (unsigned short)0;
         }
       }
     }
@@ -386,7 +404,10 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
       if (c >= NCH) break;
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        const int vt = (colb + nf * 16 + row + nk - 1) * LDR;
+        const int cl = colb + nf * 16 + row;
+        const int bo = cl >> tlog;
+        const int ul = cl & (TT - 1);
+        const int vt = (bo * VR + ul + nk - 1) * LDR;
         bf16x8x bfr;
         ((uint2*)&bfr)[0] = *(const uint2*)(lds_gz + vt + off0[c]);
         ((uint2*)&bfr)[1] = *(const uint2*)(lds_gz + vt + off1[c]);
@@ -397,7 +418,10 @@ xstage_bwd_data_kernel(const unsigned short* __restrict__ gz,
 
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      const int u = u0 + colb + nf * 16 + col;
+      const int cl = colb + nf * 16 + col;
+      const int b = bg * NB + (cl >> tlog);
+      const int u = u0 + (cl & (TT - 1));
+      if (b >= B) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int rg = row4 + r;
@@ -659,10 +683,15 @@ torch::Tensor xstage_bwd_data(torch::Tensor gz, torch::Tensor w,
   const int nk = (K + (int)stride - 1) / (int)stride;
   const int NCH = (nk + ipc - 1) / ipc;
   const int V = (256 + nk + 7) / 8 * 8 + 8;
-  const size_t lds = (size_t)(NCH * 16 * XLDW + V * 20) * 2;
   auto gx = torch::empty({B, Cin, (int)Tin}, gz.options());
   auto stream = at::hip::getCurrentHIPStream();
   const int Umax = ((int)Tin - 1 + (int)pad) / (int)stride + 1;
+  // mirror the kernel's b-fold geometry for the LDS size
+  int TT = 256;
+  while (TT >= 32 && (TT >> 1) >= Umax) TT >>= 1;
+  const int NBh = 256 / TT;
+  const int VR = ((TT + nk + 7) / 8) * 8 + 8;
+  const size_t lds = (size_t)(NCH * 16 * XLDW + NBh * VR * 20) * 2;
   const int utiles = xcdiv(Umax, 256);
   const int nz = (int)std::min<int64_t>(
       (int64_t)B * utiles,
