@@ -229,11 +229,40 @@ __global__ void im2col_bt_kernel(const unsigned short* __restrict__ x,
 // A = gz^T (Cout x B*To), Bt = k-major im2col (Cin*K x B*To); Kred =
 // B*To is huge, so grid.z slices the K loop to fill the chip.
 // ---------------------------------------------------------------------------
+// A-tile stage reading gz in its NATIVE [B][Cout][To] layout (row m =
+// co, col k = b*To + t) — kills the gzT permute+contiguous pass the
+// round-1 engine paid per weight-grad call.  16-B runs stay inside one
+// b because To % 8 == 0 (checked by the caller).
+DEV_INLINE void stage_tile_glds_gz3d(const unsigned short* __restrict__ gz,
+                                     int64_t row0, int64_t rows, int Cout,
+                                     int To, int64_t kcol0,
+                                     unsigned short* lds_base, int wid,
+                                     int lane) {
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int c = wid * 2 + i;
+    const int row = c * 16 + (lane >> 2);
+    const int byte = (lane & 3) * 16;
+    const int sbyte = swz(row, byte);
+    int64_t srow = row0 + row;          // co
+    if (srow >= rows) srow = rows - 1;
+    const int64_t k = kcol0 + sbyte / 2;  // (b, t) flat
+    const int64_t b = k / To;
+    const int64_t t = k - b * To;
+    const unsigned short* gptr =
+        gz + (b * Cout + srow) * To + t;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gptr,
+        (__attribute__((address_space(3))) void*)(lds_base + c * 512),
+        16, 0, 0);
+  }
+}
+
 __global__ void __launch_bounds__(256)
 gemm_bt_splitk_f32_kernel(const unsigned short* __restrict__ A,
                           const unsigned short* __restrict__ Bt,
                           float* __restrict__ out, int M, int64_t N,
-                          int64_t Kred, int nsplit) {
+                          int64_t Kred, int nsplit, int Cout, int To) {
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   const int TILE = GBM * GBK;
 
@@ -254,7 +283,8 @@ gemm_bt_splitk_f32_kernel(const unsigned short* __restrict__ A,
   const int ks_hi = min(ks_lo + per, ksteps);
   if (ks_lo >= ks_hi) return;
 
-  stage_tile_glds(A, m0, M, Kred, ks_lo * GBK, smem, wid, lane);
+  stage_tile_glds_gz3d(A, m0, M, Cout, To, (int64_t)ks_lo * GBK, smem,
+                       wid, lane);
   stage_tile_glds(Bt, n0, N, Kred, ks_lo * GBK, smem + TILE, wid, lane);
   __syncthreads();
 
@@ -262,7 +292,8 @@ gemm_bt_splitk_f32_kernel(const unsigned short* __restrict__ A,
   for (int ks = ks_lo; ks < ks_hi; ++ks) {
     if (ks + 1 < ks_hi) {
       unsigned short* nb = smem + (cur ^ 1) * 2 * TILE;
-      stage_tile_glds(A, m0, M, Kred, (ks + 1) * GBK, nb, wid, lane);
+      stage_tile_glds_gz3d(A, m0, M, Cout, To,
+                           (int64_t)(ks + 1) * GBK, nb, wid, lane);
       stage_tile_glds(Bt, n0, N, Kred, (ks + 1) * GBK, nb + TILE, wid,
                       lane);
     }
@@ -465,7 +496,7 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   const int NKp = (NK + 31) / 32 * 32;  // N padded only for tiling
   auto stream = at::hip::getCurrentHIPStream();
 
-  auto gzT = gz.permute({1, 0, 2}).reshape({Cout, Kred}).contiguous();
+  TORCH_CHECK(To % 8 == 0, "bwd_weight_gemm: To must be 8-aligned");
   auto btk = torch::empty({(int64_t)NKp, Kred}, x.options());
   {
     const int64_t n = (int64_t)NKp * Kred;
@@ -487,9 +518,9 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);
   dim3 grid(nt, mt, nsplit);
   hipLaunchKernelGGL(gemm_bt_splitk_f32_kernel, grid, dim3(256), lds,
-                     stream, (const unsigned short*)gzT.data_ptr(),
+                     stream, (const unsigned short*)gz.data_ptr(),
                      (const unsigned short*)btk.data_ptr(),
                      gw.data_ptr<float>(), Cout, (int64_t)NKp, Kred,
-                     nsplit);
+                     nsplit, Cout, To);
   return gw.narrow(1, 0, NK).reshape({Cout, Cin, (int64_t)K}).contiguous();
 }
