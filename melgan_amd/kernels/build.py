@@ -27,6 +27,7 @@ SOURCES = [
     "xstage.hip",
     "pqmf.hip",
     "gemm256.hip",
+    "cin1.hip",
 ]
 
 

@@ -533,6 +533,18 @@ torch::Tensor multi_l1_fwd(torch::Tensor tbl, torch::Tensor chunks,
 void multi_l1_bwd(torch::Tensor tbl, torch::Tensor chunks, int64_t nt,
                   torch::Tensor gout, torch::ScalarType dt);
 
+// Cin==1 VALU kernels (cin1.hip): the k=15 disc-input conv shapes
+bool cin1_eligible(int64_t Cin, int64_t Cout, int64_t K, int64_t stride,
+                   int64_t groups);
+torch::Tensor cin1_fwd(torch::Tensor x, torch::Tensor w,
+                       const float* bias_p, int64_t dil, int64_t pad,
+                       bool reflect, int64_t act, double slope,
+                       bool act_before, int64_t To);
+torch::Tensor cin1_bwd_data_pad(torch::Tensor gz, torch::Tensor w,
+                                int64_t dil, int64_t pad, int64_t Tin);
+void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
+                    int Tin, int pad, bool reflect, int src_len);
+
 // dedicated PQMF polyphase kernels (pqmf.hip)
 torch::Tensor pqmf_gather(torch::Tensor x, torch::Tensor fpad, int64_t KT,
                           int64_t N, int64_t pad, double scale);
@@ -574,6 +586,9 @@ torch::Tensor conv1d_fwd(torch::Tensor x, torch::Tensor w,
     bias_f = bias->to(torch::kFloat).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
+  if (cin1_eligible(Cin, Cout, K, stride, groups))
+    return cin1_fwd(x, w, bias_p, dil, pad, reflect, act, slope,
+                    act_before, To);
   if (conv1d_mfma_eligible(x, Cin, K, stride, dil, groups))
     return conv1d_fwd_mfma(x, w, bias_p, dil, pad, reflect, act, slope,
                            act_before, To);
@@ -621,6 +636,14 @@ torch::Tensor conv1d_bwd_data(torch::Tensor gz, torch::Tensor w,
                    Tpad);
     return gx;
   }
+  if (cin1_eligible(Cin, Cout, K, stride, groups)) {
+    auto gxpad = cin1_bwd_data_pad(gz, w, dil, pad, Tin);
+    if (pad == 0) return gxpad;
+    auto gx = torch::empty({B, 1, (int)Tin}, gz.options());
+    run_unpad_fold(gxpad, gx, B, (int)Tin, (int)pad, reflect,
+                   (int)Tin + 2 * (int)pad);
+    return gx;
+  }
   if (conv1d_mfma_eligible(gz, Cout, K, stride, dil, groups))
     return conv1d_bwd_data_mfma(gz, w, dil, pad, reflect, Tin);
   if (conv1d_mfma_eligible_any(gz))
@@ -651,7 +674,7 @@ torch::Tensor conv1d_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
   const int Cpg = Cin / (int)groups;
-  if (conv1d_mfma_eligible_any(x))
+  if (conv1d_mfma_eligible_any(x) && !(Cin == 1 && groups == 1))
     return conv1d_bwd_weight_mfma(gz, x, stride, dil, pad, reflect, groups,
                                   K, /*act_a=*/false,
                                   /*act_b=*/act_before, slope);
