@@ -542,6 +542,9 @@ torch::Tensor cin1_fwd(torch::Tensor x, torch::Tensor w,
                        bool act_before, int64_t To);
 torch::Tensor cin1_bwd_data_pad(torch::Tensor gz, torch::Tensor w,
                                 int64_t dil, int64_t pad, int64_t Tin);
+torch::Tensor cin1_bwd_weight(torch::Tensor gz, torch::Tensor x,
+                              int64_t K, int64_t pad, bool reflect,
+                              bool act_b, double slope);
 void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
                     int Tin, int pad, bool reflect, int src_len);
 
@@ -674,7 +677,9 @@ torch::Tensor conv1d_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
   const int Cpg = Cin / (int)groups;
-  if (conv1d_mfma_eligible_any(x) && !(Cin == 1 && groups == 1))
+  if (cin1_eligible(Cin, Cout, K, stride, groups) && dil == 1 && K <= 16)
+    return cin1_bwd_weight(gz, x, K, pad, reflect, act_before, slope);
+  if (conv1d_mfma_eligible_any(x))
     return conv1d_bwd_weight_mfma(gz, x, stride, dil, pad, reflect, groups,
                                   K, /*act_a=*/false,
                                   /*act_b=*/act_before, slope);
