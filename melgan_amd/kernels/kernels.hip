@@ -677,8 +677,9 @@ torch::Tensor conv1d_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
   const int Cin = x.size(1), Tin = x.size(2);
   const int Cpg = Cin / (int)groups;
-  if (cin1_eligible(Cin, Cout, K, stride, groups) && dil == 1 && K <= 16)
-    return cin1_bwd_weight(gz, x, K, pad, reflect, act_before, slope);
+  // NOTE: a dedicated cin1 weight-grad kernel (cin1.hip) measured 870us
+  // per call vs the mfma16 path's ~100us on the k15 shape — the 16-row
+  // MFMA tile wins on the (tiny N, huge B*T) reduction; keep mfma16.
   if (conv1d_mfma_eligible_any(x))
     return conv1d_bwd_weight_mfma(gz, x, stride, dil, pad, reflect, groups,
                                   K, /*act_a=*/false,
