@@ -496,7 +496,16 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   const int NKp = (NK + 31) / 32 * 32;  // N padded only for tiling
   auto stream = at::hip::getCurrentHIPStream();
 
-  TORCH_CHECK(To % 8 == 0, "bwd_weight_gemm: To must be 8-aligned");
+  // the 3D A-staging needs 8-aligned To (16-B k-runs must not cross a
+  // batch boundary); otherwise fall back to the permuted 2D operand,
+  // which the same stage path reads with (Cout=1, To=Kred)
+  torch::Tensor gzA = gz;
+  int a_cout = Cout, a_to = To;
+  if (To % 8 != 0) {
+    gzA = gz.permute({1, 0, 2}).reshape({Cout, Kred}).contiguous();
+    a_cout = 1;
+    a_to = (int)Kred;
+  }
   auto btk = torch::empty({(int64_t)NKp, Kred}, x.options());
   {
     const int64_t n = (int64_t)NKp * Kred;
@@ -518,9 +527,9 @@ torch::Tensor conv1d_bwd_weight_gemm(torch::Tensor gz, torch::Tensor x,
   const size_t lds = 4 * GBM * GBK * sizeof(unsigned short);
   dim3 grid(nt, mt, nsplit);
   hipLaunchKernelGGL(gemm_bt_splitk_f32_kernel, grid, dim3(256), lds,
-                     stream, (const unsigned short*)gz.data_ptr(),
+                     stream, (const unsigned short*)gzA.data_ptr(),
                      (const unsigned short*)btk.data_ptr(),
                      gw.data_ptr<float>(), Cout, (int64_t)NKp, Kred,
-                     nsplit, Cout, To);
+                     nsplit, a_cout, a_to);
   return gw.narrow(1, 0, NK).reshape({Cout, Cin, (int64_t)K}).contiguous();
 }

@@ -182,7 +182,6 @@ class FusedWNConv1dFn(torch.autograd.Function):
             if (gz.dtype == torch.bfloat16 and stride == 1 and groups == 1
                     and x.size(1) * K >= 384 and gz.size(1) >= 64
                     and (gz.size(0) * gz.size(2)) % 32 == 0
-                    and gz.size(2) % 8 == 0
                     and _os.environ.get("MELGAN_AMD_NO_GEMM", "0") != "1"):
                 gw_full = C.conv1d_bwd_weight_gemm(
                     gz, x, K, stride, dilation, pad, reflect, act_before,
@@ -370,7 +369,7 @@ class ConvTGemmFn(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             lx = torch.nn.functional.leaky_relu(x, slope) \
                 if act_before else x
-            if (x.size(0) * T) % 32 == 0 and T % 8 == 0 and \
+            if (x.size(0) * T) % 32 == 0 and \
                     gy.dtype == torch.bfloat16:
                 gw_full = C.conv1d_bwd_weight_gemm(
                     lx.contiguous(), gy, K, stride, 1, pad, False, False,
