@@ -310,9 +310,11 @@ __global__ void act_bwd_biasgrad_kernel(const T* __restrict__ gy,
   const int64_t lo = (int64_t)blockIdx.y * per;
   const int64_t hi = min(lo + per, bt_total);
   float acc = 0.f;
-  for (int64_t bt = lo + threadIdx.x; bt < hi; bt += blockDim.x) {
-    const int b = (int)(bt / Tlen);
-    const int t = (int)(bt - (int64_t)b * Tlen);
+  // carried (b, t) split: one divide per thread, not per element
+  int64_t bt = lo + threadIdx.x;
+  int b = (int)(bt / Tlen);
+  int t = (int)(bt - (int64_t)b * Tlen);
+  for (; bt < hi; bt += blockDim.x) {
     const int64_t idx = ((int64_t)b * C + c) * Tlen + t;
     float g = static_cast<float>(gy[idx]);
     float r;
@@ -326,6 +328,11 @@ __global__ void act_bwd_biasgrad_kernel(const T* __restrict__ gy,
     }
     gz[idx] = static_cast<T>(r);
     acc += r;
+    t += blockDim.x;
+    while (t >= Tlen) {
+      t -= Tlen;
+      ++b;
+    }
   }
   red[threadIdx.x] = acc;
   __syncthreads();
@@ -351,10 +358,16 @@ __global__ void bias_grad_kernel(const T* __restrict__ gz,
   const int64_t lo = (int64_t)blockIdx.y * per;
   const int64_t hi = min(lo + per, bt_total);
   float acc = 0.f;
-  for (int64_t bt = lo + threadIdx.x; bt < hi; bt += blockDim.x) {
-    const int b = (int)(bt / Tlen);
-    const int t = (int)(bt - (int64_t)b * Tlen);
+  int64_t bt = lo + threadIdx.x;
+  int b = (int)(bt / Tlen);
+  int t = (int)(bt - (int64_t)b * Tlen);
+  for (; bt < hi; bt += blockDim.x) {
     acc += static_cast<float>(gz[((int64_t)b * C + c) * Tlen + t]);
+    t += blockDim.x;
+    while (t >= Tlen) {
+      t -= Tlen;
+      ++b;
+    }
   }
   red[threadIdx.x] = acc;
   __syncthreads();
