@@ -452,9 +452,9 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
                          int Cout, int To, int K, int stride, int pad,
                          int groups, float slope, bool act_b, int KP,
                          int nsplit) {
-  constexpr int TC = 512;       // contraction t per staging round
+  constexpr int TC = 256;       // contraction t per staging round
   constexpr int TCP = TC + 8;   // gz row stride
-  constexpr int TM = 536;       // xp row stride (m in [0, TC + KP/s))
+  constexpr int TM = 280;       // xp row stride (m in [0, TC + KP/s))
   extern __shared__ __attribute__((aligned(16))) unsigned short smem[];
   unsigned short* lds_gz = smem;            // [16 co][TCP]
   unsigned short* lds_xp = smem + 16 * TCP;  // [Cpg*s][TM]
@@ -716,7 +716,7 @@ bool xstage_bwd_weight_eligible(int64_t Cin, int64_t Cout, int64_t K,
   // KP = ceil16(K) and N = Cpg*KP must be 192 (3 fragments x 4 waves)
   const int KP = (int)((K + 15) / 16 * 16);
   return Cpg * stride == 16 && Copg <= 16 && Cpg * KP == 192 && K >= 9 &&
-         KP / stride + 512 <= 536;
+         KP / stride + 256 <= 280;
 }
 
 torch::Tensor xstage_bwd_weight(torch::Tensor gz, torch::Tensor x,
@@ -730,7 +730,7 @@ torch::Tensor xstage_bwd_weight(torch::Tensor gz, torch::Tensor x,
   const int KP = (int)((K + 15) / 16 * 16);
   auto stream = at::hip::getCurrentHIPStream();
   const int gy = (int)groups * xcdiv(Copg, 16);
-  const int tchunks = xcdiv(To, 512);
+  const int tchunks = xcdiv(To, 256);
   const int64_t total = (int64_t)B * tchunks;
   int nsplit = (int)std::min<int64_t>(
       total, std::max<int64_t>(1, 768 / std::max(1, gy)));
@@ -738,7 +738,7 @@ torch::Tensor xstage_bwd_weight(torch::Tensor gz, torch::Tensor x,
   // k<K) element is written by exactly one lane, so empty is safe)
   auto gws = torch::empty({nsplit, Cout, Cpg, (int)K},
                           x.options().dtype(torch::kFloat));
-  const size_t lds = (size_t)(16 * (512 + 8) + 16 * 536) * 2;
+  const size_t lds = (size_t)(16 * (256 + 8) + 16 * 280) * 2;
   dim3 grid(1, gy, nsplit);
   hipLaunchKernelGGL(xstage_bwd_weight_kernel, grid, dim3(256), lds,
                      stream, (const unsigned short*)gz.data_ptr(),
