@@ -227,13 +227,15 @@ template <typename T>
 __global__ void unpad_fold_kernel(const T* __restrict__ gxpad,
                                   T* __restrict__ gx, int BC, int Tin,
                                   int pad, bool reflect, int src_len) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t n = (int64_t)BC * Tin;
   const int64_t step = (int64_t)gridDim.x * blockDim.x;
   const int Tpad = Tin + 2 * pad;
-  for (int64_t idx = i; idx < n; idx += step) {
-    const int64_t bc = idx / Tin;
-    const int j = (int)(idx - bc * Tin);
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n) return;
+  // carried (bc, j) split: one divide per thread, not per element
+  int64_t bc = idx / Tin;
+  int j = (int)(idx - bc * Tin);
+  for (; idx < n; idx += step) {
     const T* src = gxpad + bc * src_len;
     float v = static_cast<float>(src[pad + j]);
     if (reflect && pad > 0) {
@@ -243,6 +245,12 @@ __global__ void unpad_fold_kernel(const T* __restrict__ gxpad,
         v += static_cast<float>(src[pad + jm]);
     }
     gx[idx] = static_cast<T>(v);
+    j += (int)(step % Tin);
+    bc += step / Tin;
+    if (j >= Tin) {
+      j -= Tin;
+      ++bc;
+    }
   }
 }
 
