@@ -367,13 +367,20 @@ def test_conv1d_mfma_backward_bf16(cin, cout, k, d, pm, t):
         p.grad = None
 
 
-def test_mfma_leaky_backward_statistical():
-    """Leaky epilogue backward in bf16: mean error small; the rare
-    near-zero mask flips (bf16-vs-fp32 sign of y) and stacked bf16
-    rounding (gz, w) bound it to the ~1% class."""
-    mod = _mkconv(128, 128, 3, dilation=3, pad_mode="reflect", act="leaky")
+@pytest.mark.parametrize("cin,cout,k,d,g,s_,pm,t", [
+    (128, 128, 3, 3, 1, 1, "reflect", 1024),
+    (80, 512, 7, 1, 1, 1, "reflect", 96),       # generator input conv
+    (1024, 1024, 5, 1, 1, 1, "zero", 64),       # disc k5 (GEMM route)
+    (64, 256, 41, 4, 16, 4, "zero", 1024),      # tower L2 (xstage route)
+])
+def test_mfma_leaky_backward_statistical(cin, cout, k, d, g, s_, pm, t):
+    """Leaky epilogue backward in bf16 across the dispatch tiers: mean
+    error small; the rare near-zero mask flips (bf16-vs-fp32 sign of y)
+    and stacked bf16 rounding (gz, w) bound it to the ~1% class."""
+    mod = _mkconv(cin, cout, k, dilation=d, groups=g, stride=s_,
+                  pad_mode=pm, act="leaky")
     torch.manual_seed(2)
-    x16 = torch.randn(2, 128, 1024, device=DEV).to(torch.bfloat16)
+    x16 = torch.randn(2, cin, t, device=DEV).to(torch.bfloat16)
     from melgan_amd.ops.fused import fused_conv1d
 
     x1 = x16.clone().requires_grad_(True)
