@@ -32,6 +32,27 @@ namespace melgan {
 
 typedef __bf16 bf16x8x __attribute__((ext_vector_type(8)));
 typedef float f32x4x __attribute__((ext_vector_type(4)));
+typedef unsigned u32x2x __attribute__((ext_vector_type(2)));
+
+// two ds_read_b64 HIDDEN from hipcc's s_waitcnt bookkeeping: with the
+// glds x-prefetch in flight, hipcc cannot prove the LDS ranges disjoint
+// and would order every visible ds_read behind `s_waitcnt vmcnt(0)`
+// (22 drains in the .s), serialising the prefetch.  The caller waits
+// with an explicit lgkmcnt(0) + sched_barrier before the MFMAs
+// (guide §5.7 form iii / rule 18).
+DEV_INLINE bf16x8x ds_read_2xb64_asm(const unsigned short* p) {
+  const unsigned off = (unsigned)(unsigned long long)
+      (const __attribute__((address_space(3))) void*)p;
+  u32x2x lo, hi;
+  asm volatile("ds_read_b64 %0, %2\n\t"
+               "ds_read_b64 %1, %2 offset:8"
+               : "=v"(lo), "=v"(hi)
+               : "v"(off));
+  union { u32x2x q[2]; bf16x8x v; } r;
+  r.q[0] = lo;
+  r.q[1] = hi;
+  return r.v;
+}
 
 constexpr int XLDW = 40;   // A-image row stride (80 B, 16B-aligned rows)
 
@@ -162,6 +183,16 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
   const int XBUF = ((NB * Cpg * XROW + 511) / 512) * 512;
   const int nch_g = XBUF / 512;  // 1-KiB glds chunks per tile
 
+  // bias hoisted to registers: an ordinary global load inside the tile
+  // loop would make hipcc drain the in-flight glds (vmcnt(0)) before
+  // its use (guide §5 trap (b))
+  float bv[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int co = co_base + row4 + r;
+    bv[r] = (bias && co < co_end) ? bias[co] : 0.f;
+  }
+
   // LDS x layout: [b_off][cig][XROW]
   auto stage_sync = [&](unsigned short* dst, int bg, int t0) {
     const int gt0 = t0 * stride - pad;
@@ -241,19 +272,21 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
     for (int c = 0; c < 6; ++c) {
       if (c >= NCH) break;
+      bf16x8x bfr[4];
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
         const int cl = colb + nf * 16 + row;     // tile column
         const int bo = cl >> tlog;               // 0 when !bfold
         const int tl = cl & (TT - 1);
-        const unsigned short* p = xcur + (bo * Cpg) * XROW + xoff[c] +
-                                  tl * stride;
-        bf16x8x bfr;
-        ((uint2*)&bfr)[0] = *(const uint2*)p;
-        ((uint2*)&bfr)[1] = *(const uint2*)(p + 4);
-        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[c], bfr,
-                                                          acc[nf], 0, 0, 0);
+        bfr[nf] = ds_read_2xb64_asm(xcur + (bo * Cpg) * XROW + xoff[c] +
+                                    tl * stride);
       }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[c], bfr[nf],
+                                                          acc[nf], 0, 0, 0);
     }
 
 #pragma unroll
@@ -266,7 +299,7 @@ xstage_fwd_kernel(const unsigned short* __restrict__ x,
       for (int r = 0; r < 4; ++r) {
         const int co = co_base + row4 + r;
         if (co >= co_end) continue;
-        float o = acc[nf][r] + (bias ? bias[co] : 0.f);
+        float o = acc[nf][r] + bv[r];
         if (!act_before) o = apply_act(o, act, slope);
         union { __bf16 bb; unsigned short s; } cv{(__bf16)o};
         y[((int64_t)b * Cout + co) * To + t] = cv.s;
