@@ -214,6 +214,11 @@ class Trainer:
             fake = self.pqmf.synthesis(fake_sub) if self.multiband \
                 else fake_sub
             self.loss_stream.wait_stream(torch.cuda.current_stream())
+            # main-stream tensors consumed on the side stream: guard the
+            # allocator against reuse while side-stream work is in flight
+            fake_sub.record_stream(self.loss_stream)
+            fake.record_stream(self.loss_stream)
+            audio.record_stream(self.loss_stream)
             with torch.cuda.stream(self.loss_stream):
                 stft_loss, sub_loss = self._stft_losses(fake_sub, fake,
                                                         audio)
@@ -293,6 +298,11 @@ class Trainer:
         # ---- generator update (D still at pre-step weights) -------------
         tm.mark("g_bwd")
         g_loss.backward()
+        if adversarial and self.loss_stream is not None:
+            # the STFT-path backward nodes replayed on the loss stream;
+            # their generator-grad contributions must be complete before
+            # the fused Adam (raw-pointer tables) reads the grads
+            torch.cuda.current_stream().wait_stream(self.loss_stream)
         tm.mark("g_comm_wait")
         self.dp_g.finish()
         tm.mark("g_opt")
