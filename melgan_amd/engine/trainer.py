@@ -101,14 +101,15 @@ class Trainer:
         self.timer = PhaseTimer(enabled=bool(hp.backend.get("phase_timing"))
                                 and self.device.type == "cuda")
         self.last_phases = {}
-        # side HIP stream: the multi-resolution STFT losses (rocFFT +
-        # reductions, small grids) overlap the discriminator convs; the
-        # backward nodes replay on their forward streams, so the overlap
-        # holds in backward too (SURVEY.md §5.1 / NOTES round-2 item)
+        # side HIP stream for the multi-resolution STFT losses.  OFF BY
+        # DEFAULT: a 2000-step soak NaNs at step ~600-1600 with it on
+        # (reproduced with loss_stream alone; d_stream alone and both-off
+        # run 2000 steps clean), and record_stream + a pre-optimizer
+        # wait_stream did not cure it — root cause still open (NOTES.md).
+        # Opt in with backend: {loss_stream: true}.
         use_ls = hp.backend.get("loss_stream")
         self.loss_stream = torch.cuda.Stream() \
-            if (use_ls is None or use_ls) and self.device.type == "cuda" \
-            else None
+            if use_ls and self.device.type == "cuda" else None
         # the d-phase forward runs on its own stream so its BACKWARD
         # (replayed there by autograd) overlaps the generator backward
         use_ds = hp.backend.get("d_stream")
