@@ -44,9 +44,11 @@ DEV_INLINE bf16x8x ds_read_2xb64_asm(const unsigned short* p) {
   const unsigned off = (unsigned)(unsigned long long)
       (const __attribute__((address_space(3))) void*)p;
   u32x2x lo, hi;
+  // early-clobber: the async reads may land in the outputs before the
+  // second instruction has read the address operand (guide §5.7)
   asm volatile("ds_read_b64 %0, %2\n\t"
                "ds_read_b64 %1, %2 offset:8"
-               : "=v"(lo), "=v"(hi)
+               : "=&v"(lo), "=&v"(hi)
                : "v"(off));
   union { u32x2x q[2]; bf16x8x v; } r;
   r.q[0] = lo;
