@@ -28,8 +28,10 @@ def _worker(rank, world, port, results):
 
         model = _make_model(seed=123 + rank)  # different init per rank:
         # broadcast at engine construction must equalise them
-        engine = BucketAllReduceDP(model, bucket_cap_mb=0.05)  # tiny buckets
-        # -> multiple buckets exercise the launch-per-bucket path
+        engine = BucketAllReduceDP(model, bucket_cap_mb=0.004)  # 4 KB cap
+        # on a 15 KB model -> genuinely multiple buckets (the earlier
+        # 0.05 MB cap silently left ONE bucket), exercising the
+        # launch-per-bucket path
 
         torch.manual_seed(1000 + rank)  # different DATA per rank
         mel = torch.randn(2, 16, 8)
@@ -39,7 +41,7 @@ def _worker(rank, world, port, results):
 
         grads = torch.cat([p.grad.flatten() for p in model.parameters()])
         params = torch.cat([p.data.flatten() for p in model.parameters()])
-        results[rank] = (grads, params)
+        results[rank] = (grads, params, dict(engine.last_stats))
     finally:
         dist.destroy_process_group()
 
@@ -52,11 +54,19 @@ def test_bucketed_allreduce_matches_large_batch():
     results = mgr.dict()
     mp.spawn(_worker, args=(world, port, results), nprocs=world, join=True)
 
-    g0, p0 = results[0]
-    g1, p1 = results[1]
+    g0, p0, s0 = results[0]
+    g1, p1, _ = results[1]
     # both ranks ended with identical (averaged) gradients and identical params
     assert torch.allclose(g0, g1, atol=1e-6)
     assert torch.equal(p0, p1)
+
+    # finish() recorded the overlap artifact bench.py ships at world>1
+    assert s0["buckets"] >= 2, "0.05 MB cap should split into many buckets"
+    assert s0["bytes"] == g0.numel() * 4  # fp32 grads, every param bucketed
+    assert 0 <= s0["pre_completed"] <= s0["buckets"]
+    assert s0["finish_wait_ms"] >= 0
+    assert len(s0["bucket_ms"]) == s0["buckets"]
+    assert all(ms >= 0 for ms in s0["bucket_ms"])
 
     # single-process reference: same model (rank0 init), batch = both ranks'
     model = _make_model(seed=123)  # engine broadcast rank0 init
@@ -84,6 +94,7 @@ def test_world1_noop():
     y = model(torch.randn(1, 16, 8))
     y.mean().backward()
     engine.finish()  # no-op, must not raise
+    assert engine.last_stats == {}  # no overlap artifact at world 1
 
 
 def _gan_worker(rank, world, port, results):
