@@ -97,6 +97,63 @@ def test_world1_noop():
     assert engine.last_stats == {}  # no overlap artifact at world 1
 
 
+def _rebind_worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from melgan_amd.parallel.dp import BucketAllReduceDP
+
+        model = _make_model(seed=123)
+        engine = BucketAllReduceDP(model, bucket_cap_mb=0.004)
+        # sever the bucket binding the way an external zero_grad would:
+        # autograd then allocates FRESH grad storage and the hook must
+        # fold it into the bucket view and rebind (the _on_grad fallback)
+        for p in model.parameters():
+            p.grad = None
+        torch.manual_seed(1000 + rank)
+        mel = torch.randn(2, 16, 8)
+        model(mel).pow(2).mean().backward()
+        engine.finish()
+        rebound = all(
+            p.grad.data_ptr() == b.views[i].data_ptr()
+            for b in engine.buckets
+            for i, p in enumerate(b.params))
+        grads = torch.cat([p.grad.flatten() for p in model.parameters()])
+        results[rank] = (grads, rebound)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_grad_rebind_fallback_world2():
+    """A backward after p.grad was externally set to None still lands in
+    the buckets (fold + rebind) and all-reduces correctly."""
+    world = 2
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_rebind_worker, args=(world, 29541, results), nprocs=world,
+             join=True)
+    g0, rebound0 = results[0]
+    g1, rebound1 = results[1]
+    assert rebound0 and rebound1, "p.grad not rebound to the bucket views"
+    assert torch.allclose(g0, g1, atol=1e-6)
+
+    # reference: average of the two ranks' single-process grads
+    model = _make_model(seed=123)
+    acc = None
+    for rank in range(world):
+        torch.manual_seed(1000 + rank)
+        mel = torch.randn(2, 16, 8)
+        model.zero_grad()
+        model(mel).pow(2).mean().backward()
+        g = torch.cat([p.grad.flatten() for p in model.parameters()])
+        acc = g if acc is None else acc + g
+    assert torch.allclose(g0, acc / world, atol=1e-5)
+
+
 def _gan_worker(rank, world, port, results):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
