@@ -94,3 +94,26 @@ def test_chunked_inference_matches_full(tiny_mb_hp):
     assert chunked.shape == full.shape
     assert torch.allclose(chunked, full, atol=1e-5), \
         (chunked - full).abs().max().item()
+
+
+def test_cpu_singleband_forward_golden():
+    """Deterministic golden path (SURVEY.md §4 "numerics golden tests"):
+    the CPU composite forward is the reference every GPU kernel test
+    compares against — if it silently drifts, the whole numerics suite
+    drifts with it.  Constants were produced by this exact construction
+    on the pinned torch build."""
+    torch.manual_seed(42)
+    g = Generator(mel_channels=80, out_channels=1, base_channels=32,
+                  ratios=[8, 8, 2, 2], n_residual_layers=3).eval()
+    mel = torch.randn(1, 80, 10,
+                      generator=torch.Generator().manual_seed(7)) * 2 - 6
+    with torch.no_grad():
+        y = g(mel)
+    assert y.shape == (1, 1, 2560)  # 10 frames x 256 upsampling
+    expected_first8 = torch.tensor(
+        [0.12831259, 0.16710101, 0.08534521, 0.16874576,
+         0.11896617, 0.14558543, 0.13494481, 0.14097007])
+    assert torch.allclose(y[0, 0, :8], expected_first8, atol=1e-5), \
+        y[0, 0, :8].tolist()
+    assert abs(float(y.double().sum()) - 348.20916636) < 1e-2
+    assert abs(float(y.mean()) - 0.13601921) < 1e-5
