@@ -48,6 +48,10 @@ def main():
     ap.add_argument("-d", "--device", default=None)
     ap.add_argument("--fp32", action="store_true",
                     help="run the generator in fp32 (default bf16 on GPU)")
+    ap.add_argument("--chunk-frames", type=int, default=0,
+                    help="stream long mels in chunks of this many frames "
+                         "(overlap-discard, O(chunk) memory — SURVEY §5.7; "
+                         "0 = whole utterance at once)")
     args = ap.parse_args()
 
     hp = merge_default(HParam(args.config))
@@ -93,7 +97,11 @@ def main():
             if device.type == "cuda":
                 torch.cuda.synchronize()
             t0 = time.perf_counter()
-            audio_sub = generator.inference(mel).float()
+            if args.chunk_frames:
+                audio_sub = generator.inference_chunked(
+                    mel, chunk_frames=args.chunk_frames).float()
+            else:
+                audio_sub = generator.inference(mel).float()
             audio = pqmf.synthesis(audio_sub) if multiband else audio_sub
             if device.type == "cuda":
                 torch.cuda.synchronize()

@@ -88,7 +88,18 @@ def test_preprocess_train_infer_roundtrip(tmp_path):
     assert "TOTAL RTF" in r.stdout
     with wave.open(str(wavs[0]), "rb") as w:
         assert w.getframerate() == 22050
-        assert w.getnframes() > 10000
+        nframes_full = w.getnframes()
+        assert nframes_full > 10000
+
+    # chunked streaming inference from the CLI produces the same-length wav
+    out2 = tmp_path / "wav_out_chunked"
+    r = _run(["inference.py", "-c", str(cfg), "-p", str(cks[-1]),
+              "-i", str(meldir), "-o", str(out2), "--chunk-frames", "8"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    with wave.open(str(next(out2.glob("*.wav"))), "rb") as w:
+        # inference() pads+trims, chunked streams the raw mel — lengths
+        # agree up to the pad_frames trim margin
+        assert abs(w.getnframes() - nframes_full) <= 2 * 10 * 256
 
 
 @pytest.mark.timeout(300)
