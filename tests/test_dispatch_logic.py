@@ -107,6 +107,29 @@ def test_convt_a2_fold_matches_bwd_data(r):
     assert (gx - x.grad).abs().max() < 1e-5
 
 
+@pytest.mark.parametrize("r", [2, 4, 8])
+def test_convt_weightgrad_roleswap_identity(r):
+    """Third leg of the convT GEMM route: the weight gradient equals the
+    ROLE-SWAPPED conv1d weight-grad — treat gy as the conv input and the
+    (pre-activated) x as the grad_output, same stride/pad as bwd-data.
+    This is what conv1d_bwd_weight_gemm computes on the split-K engine."""
+    import torch.nn.functional as F
+
+    cin, cout, K = 10, 5, 2 * r
+    p = r // 2
+    torch.manual_seed(0)
+    x = torch.randn(2, cin, 13)
+    w = torch.randn(cin, cout, K, requires_grad=True)
+    y = F.conv_transpose1d(x, w, None, stride=r, padding=p)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    gw = torch.nn.grad.conv1d_weight(gy, (cin, cout, K), x,
+                                     stride=(r,), padding=(p,))
+    assert (gw - w.grad).abs().max() < 1e-4, \
+        (gw - w.grad).abs().max().item()
+
+
 def test_convt_gemm_eligibility():
     mod8 = _convt_mod(512, 256, 8)
     mod_pqmf = _convt_mod(4, 1, 4)  # PQMF synthesis shape: stays off GEMM
