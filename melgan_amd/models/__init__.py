@@ -7,6 +7,16 @@ from melgan_amd.models.discriminator import (  # noqa: F401
 
 def build_generator(hp):
     """Construct a Generator from an HParam tree (reference YAML schema)."""
+    prod = 1
+    for r in hp.model.generator_ratio:
+        prod *= r
+    if prod * hp.model.out_channels != hp.audio.hop_length:
+        raise ValueError(
+            f"model geometry mismatch: prod(generator_ratio)={prod} x "
+            f"out_channels={hp.model.out_channels} must equal "
+            f"audio.hop_length={hp.audio.hop_length} — the generator has "
+            "to upsample mel frames back to waveform samples (otherwise "
+            "training fails with a cryptic STFT-loss shape error)")
     return Generator(
         mel_channels=hp.audio.n_mel_channels,
         out_channels=hp.model.out_channels,

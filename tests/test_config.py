@@ -52,3 +52,15 @@ log: {chkpt_dir: c, log_dir: l}
     hp = merge_default(HParam(raw=raw))
     assert hp.backend.use_hip_kernels is True
     assert hp.model.generator_ratio is not None
+
+
+def test_geometry_mismatch_raises(tiny_hp):
+    """prod(ratios) x out_channels != hop_length must fail LOUDLY at
+    model construction, not deep inside the STFT loss."""
+    import pytest
+
+    from melgan_amd.models import build_generator
+
+    tiny_hp.model.generator_ratio = [8, 8, 2]  # 128 x bands != hop 256
+    with pytest.raises(ValueError, match="geometry mismatch"):
+        build_generator(tiny_hp)
