@@ -100,49 +100,10 @@ cin1_bwd_data_kernel(const T* __restrict__ gz, const T* __restrict__ w,
   }
 }
 
-// gw[co][0][k] = sum_{b,t} gz[b][co][t] * pre(x[b][0][t + k - pad])
-// block = (b, t-chunk); wave w owns co = w, w+4, ...; lane-strided t
-// with K fp32 register accumulators; shfl reduce + one atomicAdd per
-// (wave, co, k).  stride 1, K <= 16.
-template <typename T>
-__global__ void __launch_bounds__(256)
-cin1_bwd_weight_kernel(const T* __restrict__ gz, const T* __restrict__ x,
-                       float* __restrict__ gw, int B, int Tin, int Cout,
-                       int To, int K, int pad, bool reflect, float slope,
-                       bool act_b, int CH) {
-  extern __shared__ __attribute__((aligned(16))) float smw[];  // [CH+K-1]
-  const int t0 = blockIdx.x * CH;
-  const int b = blockIdx.y;
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int span = CH + K - 1;
-  const T* xr = x + (int64_t)b * Tin;
-  for (int i = tid; i < span; i += 256) {
-    float v = read_pad(xr, t0 + i - pad, Tin, reflect);
-    smw[i] = leaky_pre(v, slope, act_b);
-  }
-  __syncthreads();
-  const int t_hi = min(t0 + CH, To);
-  for (int co = wid; co < Cout; co += 4) {
-    const T* zr = gz + ((int64_t)b * Cout + co) * To;
-    float acc[16] = {};
-    for (int t = t0 + lane; t < t_hi; t += 64) {
-      const float gzv = static_cast<float>(zr[t]);
-      const float* xs = smw + (t - t0);
-#pragma unroll
-      for (int k = 0; k < 16; ++k)
-        if (k < K) acc[k] = fmaf(gzv, xs[k], acc[k]);
-    }
-#pragma unroll
-    for (int k = 0; k < 16; ++k) {
-      if (k >= K) break;
-      float v = acc[k];
-      for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
-      if (lane == 0) atomicAdd(&gw[co * K + k], v);
-    }
-  }
-}
+// (a dedicated cin1 weight-grad kernel lived here; it measured 870us
+// per call vs ~100us for the mfma16 path on the k15 shape — the 16-row
+// MFMA tile wins the (tiny N, huge B*T) reduction — so it was removed.
+// See the dispatch note in conv1d_bwd_weight, kernels.hip.)
 
 }  // namespace melgan
 
@@ -204,24 +165,3 @@ torch::Tensor cin1_bwd_data_pad(torch::Tensor gz, torch::Tensor w,
   return gxpad;
 }
 
-torch::Tensor cin1_bwd_weight(torch::Tensor gz, torch::Tensor x,
-                              int64_t K, int64_t pad, bool reflect,
-                              bool act_b, double slope) {
-  const int B = gz.size(0), Cout = gz.size(1), To = gz.size(2);
-  const int Tin = x.size(2);
-  auto gw = torch::zeros({Cout, 1, (int)K},
-                         x.options().dtype(torch::kFloat));
-  const int CH = 2048;
-  const size_t lds = (size_t)(CH + K - 1) * 4;
-  auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(c1cdiv(To, CH), B);
-  AT_DISPATCH_FLOATING_TYPES_AND2(
-      at::kBFloat16, at::kHalf, gz.scalar_type(), "cin1_bwd_weight", [&] {
-        hipLaunchKernelGGL((cin1_bwd_weight_kernel<scalar_t>), grid,
-                           dim3(256), lds, stream,
-                           gz.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                           gw.data_ptr<float>(), B, Tin, Cout, To, (int)K,
-                           (int)pad, reflect, (float)slope, act_b, CH);
-      });
-  return gw;
-}

@@ -555,9 +555,6 @@ torch::Tensor cin1_fwd(torch::Tensor x, torch::Tensor w,
                        bool act_before, int64_t To);
 torch::Tensor cin1_bwd_data_pad(torch::Tensor gz, torch::Tensor w,
                                 int64_t dil, int64_t pad, int64_t Tin);
-torch::Tensor cin1_bwd_weight(torch::Tensor gz, torch::Tensor x,
-                              int64_t K, int64_t pad, bool reflect,
-                              bool act_b, double slope);
 void run_unpad_fold(torch::Tensor gxpad, torch::Tensor gx, int BC,
                     int Tin, int pad, bool reflect, int src_len);
 
