@@ -4,7 +4,7 @@
 //   conv1d_fwd / conv1d_bwd_data / conv1d_bwd_weight   (K1, K3, K9-analysis)
 //   conv_transpose1d_fwd / _bwd_data / _bwd_weight     (K2, K9-synthesis)
 //   act_bwd, leaky_bwd_inplace, conv1d_bias_grad       (K6, K7 epilogues)
-//   multi_tensor_adam                                  (K12)
+//   multi_tensor_adam_tbl                              (K12)
 //
 // Design notes (per /opt/skills/guides/cdna_hip_programming.md):
 //   - wave64: blocks are multiples of 64; the t (time) axis maps to lanes
@@ -931,47 +931,9 @@ std::vector<torch::Tensor> wn_bwd(torch::Tensor gw, torch::Tensor v,
   return {gv, gg};
 }
 
-void multi_tensor_adam(std::vector<torch::Tensor> params,
-                       std::vector<torch::Tensor> grads,
-                       std::vector<torch::Tensor> exp_avgs,
-                       std::vector<torch::Tensor> exp_avg_sqs, double lr,
-                       double beta1, double beta2, double eps, double wd,
-                       int64_t step) {
-  const int nt = (int)params.size();
-  TORCH_CHECK(nt > 0);
-  // host-side tables: [p*nt | g*nt | m*nt | v*nt | numel*nt] as int64,
-  // plus the chunk -> (tensor, offset) map; one H2D copy per step.
-  std::vector<int64_t> tbl(nt * 5);
-  std::vector<int64_t> chunks;  // packed (tensor_idx << 48) | offset
-  for (int i = 0; i < nt; ++i) {
-    TORCH_CHECK(params[i].scalar_type() == torch::kFloat,
-                "fused adam: fp32 master params only");
-    tbl[0 * nt + i] = (int64_t)params[i].data_ptr();
-    tbl[1 * nt + i] = (int64_t)grads[i].data_ptr();
-    tbl[2 * nt + i] = (int64_t)exp_avgs[i].data_ptr();
-    tbl[3 * nt + i] = (int64_t)exp_avg_sqs[i].data_ptr();
-    tbl[4 * nt + i] = params[i].numel();
-    for (int64_t off = 0; off < params[i].numel(); off += ADAM_CHUNK)
-      chunks.push_back(((int64_t)i << 48) | off);
-  }
-  auto opts = torch::TensorOptions().dtype(torch::kLong);
-  auto dev = params[0].device();
-  auto tbl_d = torch::from_blob(tbl.data(), {(int64_t)tbl.size()}, opts)
-                   .to(dev, /*non_blocking=*/false);
-  auto chunks_d =
-      torch::from_blob(chunks.data(), {(int64_t)chunks.size()}, opts)
-          .to(dev, false);
-  float bc1 = 1.f - powf((float)beta1, (float)step);
-  float bc2 = 1.f - powf((float)beta2, (float)step);
-  auto stream = at::hip::getCurrentHIPStream();
-  int64_t* tp = tbl_d.data_ptr<int64_t>();
-  hipLaunchKernelGGL(multi_tensor_adam_kernel,
-                     dim3((int)chunks.size()), dim3(256), 0, stream,
-                     (float**)tp, (float**)(tp + nt), (float**)(tp + 2 * nt),
-                     (float**)(tp + 3 * nt), tp + 4 * nt,
-                     chunks_d.data_ptr<int64_t>(), (float)lr, (float)beta1,
-                     (float)beta2, (float)eps, (float)wd, bc1, bc2);
-}
+// (a per-step-H2D multi_tensor_adam variant lived here; superseded by
+// multi_tensor_adam_tbl below, which caches the device tables across
+// steps — zero per-step H2D, hipGraph-capture safe.  See ops/adam.py.)
 
 void multi_tensor_adam_tbl(torch::Tensor tbl, torch::Tensor chunks,
                            int64_t nt, double lr, double beta1, double beta2,
@@ -1018,6 +980,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_bwd_weight_gemm", &conv1d_bwd_weight_gemm);
   m.def("pqmf_gather", &pqmf_gather);
   m.def("pqmf_scatter", &pqmf_scatter);
-  m.def("multi_tensor_adam", &multi_tensor_adam);
   m.def("multi_tensor_adam_tbl", &multi_tensor_adam_tbl);
 }
