@@ -30,7 +30,7 @@ from __future__ import annotations
 import torch
 
 from melgan_amd.ops import _ext
-from melgan_amd.ops.wnconv import ACT_LEAKY, ACT_NONE, ACT_TANH
+from melgan_amd.ops.wnconv import ACT_NONE
 
 
 def _wn_fold(v: torch.Tensor, g: torch.Tensor, dtype: torch.dtype):
