@@ -155,6 +155,39 @@ def _default_ratios(total: int):
     return ratios
 
 
+# keys the framework reads (top two levels); used for typo warnings only —
+# unknown keys are reported, never rejected (reference configs may carry
+# extra family keys)
+_KNOWN_KEYS = {
+    "data": {"train_dir", "val_dir"},
+    "train": {"batch_size", "num_workers", "seed", "max_steps",
+              "discriminator_train_start_steps", "adam", "stft_loss_params",
+              "lambda_adv"},
+    "audio": {"n_mel_channels", "sampling_rate", "hop_length",
+              "filter_length", "win_length", "segment_length", "pad_short",
+              "mel_fmin", "mel_fmax"},
+    "model": {"out_channels", "base_channels", "n_residual_layers",
+              "generator_ratio", "num_D", "ndf", "n_layers_D",
+              "downsamp_factor", "feat_match", "use_subband_stft_loss",
+              "pqmf_cutoff"},
+    "log": {"summary_interval", "validation_interval", "save_interval",
+            "chkpt_dir", "log_dir"},
+    "backend": {"use_hip_kernels", "amp", "bucket_cap_mb", "fused_adam",
+                "loss_stream", "d_stream", "debug_sync", "phase_timing"},
+}
+
+
+def unknown_keys(hp) -> list[str]:
+    """Dotted paths of config keys the framework never reads (typo check)."""
+    out = []
+    for g, sub in hp.items():
+        if g not in _KNOWN_KEYS:
+            out.append(g)
+        elif isinstance(sub, dict):
+            out.extend(f"{g}.{k}" for k in sub if k not in _KNOWN_KEYS[g])
+    return out
+
+
 def check_hp_compat(old_hp_str: str, new_hp_str: str, logger=None) -> bool:
     """Family behaviour on resume: warn (not fail) when hp_str differs."""
     same = old_hp_str == new_hp_str

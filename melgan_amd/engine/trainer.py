@@ -94,6 +94,13 @@ class Trainer:
         self.use_amp = (self.amp_dtype is not None
                         and self.device.type == "cuda")
         self.logger = get_logger(hp.log.log_dir, name) if rank == 0 else None
+        if self.logger:
+            from melgan_amd.config import unknown_keys
+
+            unk = unknown_keys(hp)
+            if unk:
+                self.logger.warning(
+                    f"config keys the framework never reads (typo?): {unk}")
         self.writer = MetricsWriter(hp.log.log_dir, name,
                                     enable_tb=enable_tb, rank=rank)
         self.lambda_adv = hp.train.get("lambda_adv") or 2.5

@@ -64,3 +64,28 @@ def test_geometry_mismatch_raises(tiny_hp):
     tiny_hp.model.generator_ratio = [8, 8, 2]  # 128 x bands != hop 256
     with pytest.raises(ValueError, match="geometry mismatch"):
         build_generator(tiny_hp)
+
+
+def test_unknown_key_detection():
+    """§5.6: unknown keys warn (typo check), never reject."""
+    from melgan_amd.config import HParam, merge_default, unknown_keys
+
+    raw = """
+audio: {segment_lenght: 8192, hop_length: 256}
+trian: {batch_size: 4}
+model: {out_channels: 4}
+"""
+    hp = merge_default(HParam(raw=raw))
+    unk = unknown_keys(hp)
+    assert "audio.segment_lenght" in unk
+    assert "trian" in unk
+    assert not any(u.startswith("model") for u in unk)
+
+    # every shipped config is clean
+    import glob
+    import os
+
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for cfg in glob.glob(os.path.join(here, "configs", "*.yaml")):
+        hp = merge_default(HParam(cfg))
+        assert unknown_keys(hp) == [], (cfg, unknown_keys(hp))
