@@ -409,7 +409,12 @@ class Trainer:
             audio = audio.to(self.device)
             fake_sub = self.generator(mel).float()
             fake = self.pqmf.synthesis(fake_sub) if self.multiband else fake_sub
-            sc, mag = self.stft_full(fake.squeeze(1), audio.squeeze(1).float())
+            # full-length val clips: the synthesised length (mel_frames*hop)
+            # overshoots the true audio length (mel framing rounds up) —
+            # compare on the common prefix
+            n_s = min(fake.size(2), audio.size(2))
+            sc, mag = self.stft_full(fake[:, :, :n_s].squeeze(1),
+                                     audio[:, :, :n_s].squeeze(1).float())
             tot_sc += float(sc)
             tot_mag += float(mag)
             n += 1

@@ -114,3 +114,87 @@ def test_distributed_sampler_sharding(tiny_hp):
     # each rank sees ~half (DistributedSampler pads to equal shards)
     assert abs(n0 - n1) <= tiny_hp.train.batch_size
     assert n0 + n1 >= total - 2 * tiny_hp.train.batch_size
+
+
+def test_validate_full_length_disk_clips(tmp_path):
+    """validate() on a REAL on-disk val set: full-length uncropped clips of
+    varying odd lengths. The synthesised audio (mel_frames*hop samples)
+    overshoots the true clip length — validate must compare on the common
+    prefix instead of crashing with an STFT shape mismatch, and the val
+    loader must use batch_size 1 (clips can't be collated)."""
+    import wave
+
+    import numpy as np
+
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.data.dataset import create_dataloader
+
+    hop = 256
+    for name, n in [("a", 20000), ("b", 33333)]:  # odd, non-multiple-of-hop
+        sig = 0.3 * np.sin(2 * np.pi * 220 * np.arange(n) / 22050)
+        pcm = np.clip(sig * 32767, -32768, 32767).astype(np.int16)
+        with wave.open(str(tmp_path / f"{name}.wav"), "wb") as w:
+            w.setnchannels(1)
+            w.setsampwidth(2)
+            w.setframerate(22050)
+            w.writeframes(pcm.tobytes())
+        # preprocess-style framing: one extra frame past n//hop
+        torch.save(torch.randn(80, n // hop + 1) * 2 - 6,
+                   str(tmp_path / f"{name}.mel"))
+    raw = f"""
+data: {{train_dir: {tmp_path}, val_dir: {tmp_path}}}
+train: {{batch_size: 4, num_workers: 0, discriminator_train_start_steps: 0}}
+audio: {{segment_length: 4096, hop_length: {hop}}}
+model: {{out_channels: 4, base_channels: 64, generator_ratio: [8, 4, 2],
+         use_subband_stft_loss: true}}
+log: {{chkpt_dir: {tmp_path}/ck, log_dir: {tmp_path}/logs}}
+backend: {{amp: none, fused_adam: false}}
+"""
+    hp = merge_default(HParam(raw=raw))
+    tr = Trainer(hp, "val_disk", enable_tb=False)
+    vl = create_dataloader(hp, train=False)
+    assert vl.batch_size == 1  # variable-length clips can't be collated
+    sc, mag = tr.validate(vl)
+    assert sc > 0 and mag > 0
+
+
+def test_epoch_loop_refreshes_dataset(tiny_hp, tmp_path):
+    """train() re-iterates the DataLoader every epoch (fresh random crops)
+    instead of memoising the first epoch's batches, and calls
+    DistributedSampler.set_epoch so shards reshuffle."""
+    from torch.utils.data import DataLoader, Dataset
+    from torch.utils.data.distributed import DistributedSampler
+
+    class Counting(Dataset):
+        calls = 0
+
+        def __len__(self):
+            return 4
+
+        def __getitem__(self, i):
+            Counting.calls += 1
+            frames = tiny_hp.audio.segment_length // tiny_hp.audio.hop_length
+            g = torch.Generator().manual_seed(i + Counting.calls)
+            return (torch.randn(80, frames, generator=g),
+                    torch.rand(1, tiny_hp.audio.segment_length,
+                               generator=g) * 1.6 - 0.8)
+
+    epochs_seen = []
+
+    class Sampler(DistributedSampler):
+        def set_epoch(self, epoch):
+            epochs_seen.append(epoch)
+            super().set_epoch(epoch)
+
+    tiny_hp.log.chkpt_dir = str(tmp_path / "ck")
+    tiny_hp.log.log_dir = str(tmp_path / "logs")
+    tiny_hp.log.save_interval = 1000
+    ds = Counting()
+    dl = DataLoader(ds, batch_size=2,
+                    sampler=Sampler(ds, num_replicas=1, rank=0))
+    tr = Trainer(tiny_hp, "epochs", enable_tb=False)
+    end = tr.train(dl, max_steps=4)  # 2 batches/epoch -> 2 epochs
+    assert end == 4
+    # every step loaded FRESH items (4 steps x batch 2 = 8 loads)
+    assert Counting.calls >= 8
+    assert epochs_seen[:2] == [0, 1]
