@@ -29,7 +29,7 @@ audio:
   win_length: 1024
   sampling_rate: 22050
 model: {{out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}}
-log: {{summary_interval: 2, validation_interval: 1000, save_interval: 3,
+log: {{summary_interval: 2, validation_interval: 2, save_interval: 3,
       chkpt_dir: {ck}, log_dir: {lg}}}
 backend: {{amp: none, fused_adam: false}}
 """
@@ -68,11 +68,21 @@ def test_preprocess_train_infer_roundtrip(tmp_path):
     mels = list(data.glob("*.mel"))
     assert len(mels) == 3
 
-    # train 3 steps on the real (tiny) dataset; checkpoint at step 3
+    # train 3 steps on the real (tiny) dataset; checkpoint at step 3.
+    # validation_interval=2 makes validate() run mid-train on the REAL
+    # val set (full-length uncropped clips — regression for the
+    # synthesised-length overshoot crash)
     r = _run(["train.py", "-c", str(cfg), "-n", "e2e", "--max-steps", "3"])
     assert r.returncode == 0, r.stderr[-2000:]
     cks = sorted((tmp_path / "ck").glob("e2e_*.pt"))
     assert cks, "no checkpoint written"
+
+    # family resume contract: train.py -p continues from the checkpoint
+    r = _run(["train.py", "-c", str(cfg), "-n", "e2e_resume",
+              "-p", str(cks[-1]), "--max-steps", "6"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert sorted((tmp_path / "ck").glob("e2e_resume_0006.pt")), \
+        "resume did not continue to step 6"
 
     # inference: mel .npy -> wav with RTF report
     meldir = tmp_path / "mels"
