@@ -156,3 +156,23 @@ def test_bench_gpus2_reexecs_torchrun(tmp_path):
     ov = rec["config"]["dp_overlap"]
     assert ov["g"]["buckets"] >= 1 and ov["g"]["bytes"] > 0
     assert len(ov["g"]["bucket_ms"]) == ov["g"]["buckets"]
+
+
+@pytest.mark.timeout(600)
+def test_train_cli_world2_gloo(tmp_path):
+    """train.py under torchrun with 2 ranks (gloo on CPU): rendezvous on
+    127.0.0.1, per-rank sharding, rank-0-only checkpoint/metrics — the
+    same path the driver's multi-GPU run takes over RCCL."""
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text(TINY_CFG.format(d="null", ck=str(tmp_path / "ck"),
+                                   lg=str(tmp_path / "lg")))
+    r = _run(["-m", "torch.distributed.run", "--nnodes=1",
+              "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+              "--master-port", "29517", "train.py", "-c", str(cfg),
+              "-n", "dist2", "--synthetic", "--max-steps", "4"])
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    cks = sorted((tmp_path / "ck").glob("dist2_*.pt"))
+    assert cks, "rank 0 wrote no checkpoint"
+    # exactly ONE metrics stream (rank 0); rank 1 must not write
+    metrics = list((tmp_path / "lg").glob("dist2_metrics.jsonl"))
+    assert len(metrics) == 1
