@@ -163,12 +163,17 @@ def test_train_cli_world2_gloo(tmp_path):
     """train.py under torchrun with 2 ranks (gloo on CPU): rendezvous on
     127.0.0.1, per-rank sharding, rank-0-only checkpoint/metrics — the
     same path the driver's multi-GPU run takes over RCCL."""
+    import socket
+
     cfg = tmp_path / "cfg.yaml"
     cfg.write_text(TINY_CFG.format(d="null", ck=str(tmp_path / "ck"),
                                    lg=str(tmp_path / "lg")))
+    with socket.socket() as s:  # free rendezvous port (xdist-safe)
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
     r = _run(["-m", "torch.distributed.run", "--nnodes=1",
               "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-              "--master-port", "29517", "train.py", "-c", str(cfg),
+              "--master-port", str(port), "train.py", "-c", str(cfg),
               "-n", "dist2", "--synthetic", "--max-steps", "4"])
     assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
     cks = sorted((tmp_path / "ck").glob("dist2_*.pt"))
