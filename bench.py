@@ -188,6 +188,20 @@ def main():
             out["config"]["dp_overlap"] = comm
         print(json.dumps(out))
 
+    # Success path ends HERE with an explicit clean exit: interpreter/HIP
+    # teardown after a full GPU training run can segfault (observed rc 139
+    # AFTER all output, ROCm 7.2 — see NOTES.md "exit-teardown segfault"),
+    # and under torchrun a rank dying at teardown turns a finished bench
+    # into a reported failure.  Flush, tear down the process group, and
+    # exit without running the hazardous destructor chain.
+    import sys
+
+    if world > 1 and torch.distributed.is_initialized():
+        torch.distributed.destroy_process_group()
+    sys.stdout.flush()
+    sys.stderr.flush()
+    os._exit(0)
+
 
 def measure_rtf(hp, device, n_frames: int = 860, reps: int = 20,
                 dtype=torch.bfloat16):
