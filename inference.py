@@ -118,6 +118,14 @@ def main():
     print(f"TOTAL RTF: {total_gen_s/total_audio_s:.5f} "
           f"({total_audio_s/total_gen_s:.1f}x realtime) @ {sr} Hz")
 
+    # explicit clean exit on success (HIP teardown segfault after GPU
+    # work — NOTES.md "exit-teardown segfault")
+    import sys
+
+    sys.stdout.flush()
+    sys.stderr.flush()
+    os._exit(0)
+
 
 if __name__ == "__main__":
     main()

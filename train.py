@@ -59,6 +59,20 @@ def main():
     trainer.train(trainloader, valloader, max_steps=args.max_steps)
     trainer.save()
 
+    # explicit clean exit on success: interpreter/HIP teardown after a GPU
+    # training run can segfault (rc 139) AFTER all work completed — see
+    # NOTES.md "exit-teardown segfault"; under torchrun that turns a
+    # finished run into a reported failure
+    import sys
+
+    if world > 1 and torch.distributed.is_initialized():
+        torch.distributed.destroy_process_group()
+    sys.stdout.flush()
+    sys.stderr.flush()
+    import os
+
+    os._exit(0)
+
 
 if __name__ == "__main__":
     main()
