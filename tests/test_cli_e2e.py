@@ -111,6 +111,17 @@ def test_preprocess_train_infer_roundtrip(tmp_path):
         # agree up to the pad_frames trim margin
         assert abs(w.getnframes() - nframes_full) <= 2 * 10 * 256
 
+    # .mel input files (preprocess output) work directly too
+    meldir2 = tmp_path / "mels_raw"
+    meldir2.mkdir()
+    (meldir2 / "utt0.mel").write_bytes(mels[0].read_bytes())
+    out3 = tmp_path / "wav_out_mel"
+    r = _run(["inference.py", "-c", str(cfg), "-p", str(cks[-1]),
+              "-i", str(meldir2), "-o", str(out3)])
+    assert r.returncode == 0, r.stderr[-2000:]
+    with wave.open(str(next(out3.glob("*.wav"))), "rb") as w:
+        assert w.getnframes() == nframes_full  # same mel -> same length
+
 
 @pytest.mark.timeout(300)
 def test_bench_cpu_contract(tmp_path):
