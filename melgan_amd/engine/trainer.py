@@ -94,6 +94,10 @@ class Trainer:
         self.use_amp = (self.amp_dtype is not None
                         and self.device.type == "cuda")
         self.logger = get_logger(hp.log.log_dir, name) if rank == 0 else None
+        if self.amp_dtype is torch.float16 and self.logger:
+            self.logger.warning(
+                "backend.amp=fp16 compiles but is UNTESTED on GPU (no loss "
+                "scaling; bf16 is the production dtype — NOTES.md)")
         if self.logger:
             from melgan_amd.config import unknown_keys
 
