@@ -89,3 +89,23 @@ model: {out_channels: 4}
     for cfg in glob.glob(os.path.join(here, "configs", "*.yaml")):
         hp = merge_default(HParam(cfg))
         assert unknown_keys(hp) == [], (cfg, unknown_keys(hp))
+
+
+def test_hparam_pickles_into_dataloader_workers():
+    """num_workers>0 pickles the dataset (hp inside) into worker
+    processes — HParam/Dotdict must survive the round trip with hp_str."""
+    import pickle
+
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.data.dataset import create_dataloader
+
+    hp = merge_default(HParam(raw="train: {batch_size: 2, num_workers: 2}\n"
+                                  "audio: {segment_length: 4096, "
+                                  "hop_length: 256}"))
+    hp2 = pickle.loads(pickle.dumps(hp))
+    assert hp2.train.batch_size == 2
+    assert hp2.hp_str == hp.hp_str
+    dl = create_dataloader(hp, train=True, synthetic=True)
+    assert dl.num_workers == 2
+    mel, audio = next(iter(dl))
+    assert mel.shape == (2, 80, 16) and audio.shape == (2, 1, 4096)
