@@ -907,3 +907,45 @@ backend: {amp: fp16, fused_adam: true}
     for _ in range(3):
         gl, parts, dl = tr.train_step(mel, audio)
     assert torch.isfinite(gl) and torch.isfinite(dl)
+
+
+def test_validate_full_length_clips_gpu():
+    """validate() on full-length clips whose true length != mel_frames*hop
+    (real-corpus shape): the common-prefix trim must feed the fused STFT
+    pipeline odd lengths without error, and return finite losses."""
+    from torch.utils.data import DataLoader, Dataset
+
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.engine.trainer import Trainer
+
+    raw = """
+data: {train_dir: null, val_dir: null}
+train: {batch_size: 4, num_workers: 0, discriminator_train_start_steps: 0}
+audio: {segment_length: 8192, hop_length: 256}
+model: {out_channels: 4, base_channels: 128, generator_ratio: [8, 4, 2],
+        use_subband_stft_loss: true}
+log: {chkpt_dir: /tmp/gpu_val_ck, log_dir: /tmp/gpu_val_logs}
+backend: {amp: bf16, fused_adam: true}
+"""
+    hp = merge_default(HParam(raw=raw))
+    tr = Trainer(hp, "gpu_val", device=torch.device(DEV), enable_tb=False)
+
+    class FullClips(Dataset):
+        lengths = [20000, 33333]  # odd, non-multiple-of-hop
+
+        def __len__(self):
+            return 2
+
+        def __getitem__(self, i):
+            n = self.lengths[i]
+            g = torch.Generator().manual_seed(i)
+            mel = torch.randn(80, n // 256 + 1, generator=g) * 2 - 6
+            audio = torch.rand(1, n, generator=g) * 1.6 - 0.8
+            return mel, audio
+
+    vl = DataLoader(FullClips(), batch_size=1)
+    sc, mag = tr.validate(vl)
+    assert sc > 0 and mag > 0
+    import math
+
+    assert math.isfinite(sc) and math.isfinite(mag)
