@@ -209,3 +209,76 @@ def test_full_gan_step_world2():
     g1, d1 = results[1]
     assert torch.equal(g0, g1), "generator params diverged across ranks"
     assert torch.equal(d0, d1), "discriminator params diverged across ranks"
+
+
+def _resume_worker(rank, world, port, ck_path, log_dir, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from melgan_amd.config import HParam, merge_default
+        from melgan_amd.engine.trainer import Trainer
+
+        raw = f"""
+data: {{train_dir: null, val_dir: null}}
+train: {{batch_size: 2, num_workers: 0, discriminator_train_start_steps: 0}}
+audio: {{segment_length: 4096, hop_length: 256}}
+model: {{out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}}
+log: {{chkpt_dir: {log_dir}/ck2, log_dir: {log_dir}/lg{rank}}}
+backend: {{amp: none, fused_adam: false}}
+"""
+        hp = merge_default(HParam(raw=raw))
+        torch.manual_seed(500 + rank)  # rank-varying init: load must win
+        tr = Trainer(hp, f"resume{rank}", rank=rank, world_size=world,
+                     device=torch.device("cpu"), enable_tb=False)
+        tr.load(ck_path)
+        step0 = tr.step
+        torch.manual_seed(2000 + rank)  # rank-specific data
+        mel = torch.randn(2, 80, 16)
+        audio = torch.rand(2, 1, 4096) * 1.6 - 0.8
+        tr.train_step(mel, audio)
+        params = torch.cat([p.data.flatten()
+                            for p in tr.generator.parameters()])
+        results[rank] = (step0, params)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_resume_scale_out_world2(tmp_path):
+    """Production scale-out: train single-process, checkpoint, resume on
+    TWO ranks — both ranks restore the same state (broadcast at engine
+    construction must not clobber the loaded weights) and stay in
+    lockstep after a DP step on different data."""
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.engine.trainer import Trainer
+
+    raw = f"""
+data: {{train_dir: null, val_dir: null}}
+train: {{batch_size: 2, num_workers: 0, discriminator_train_start_steps: 0}}
+audio: {{segment_length: 4096, hop_length: 256}}
+model: {{out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}}
+log: {{chkpt_dir: {tmp_path}/ck, log_dir: {tmp_path}/lg}}
+backend: {{amp: none, fused_adam: false}}
+"""
+    hp = merge_default(HParam(raw=raw))
+    torch.manual_seed(0)
+    tr = Trainer(hp, "seed_run", device=torch.device("cpu"),
+                 enable_tb=False)
+    mel = torch.randn(2, 80, 16)
+    audio = torch.rand(2, 1, 4096) * 1.6 - 0.8
+    tr.train_step(mel, audio)
+    tr.train_step(mel, audio)
+    ck = tr.save()
+    assert ck
+
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_resume_worker, args=(2, 29533, ck, str(tmp_path), results),
+             nprocs=2, join=True)
+    s0, p0 = results[0]
+    s1, p1 = results[1]
+    assert s0 == 2 and s1 == 2  # step counter resumed on both ranks
+    assert torch.equal(p0, p1)  # ranks in lockstep after the DP step
