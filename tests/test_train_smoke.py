@@ -231,3 +231,26 @@ backend: {{amp: none, fused_adam: false}}
         _, parts, _ = tr.train_step(mel, audio)
         losses.append(float(parts["stft"].detach()))
     assert losses[-1] < 0.8 * losses[0], (losses[0], losses[-1])
+
+
+def test_seeded_run_reproducible(tiny_mb_hp, tmp_path):
+    """Two fresh trainers with the same seed produce bit-identical loss
+    streams on CPU — no hidden nondeterminism (dict ordering, uncached
+    RNG draws) in the step."""
+    tiny_mb_hp.log.chkpt_dir = str(tmp_path / "ck")
+    tiny_mb_hp.log.log_dir = str(tmp_path / "logs")
+
+    def run():
+        torch.manual_seed(77)
+        tr = Trainer(tiny_mb_hp, "det", enable_tb=False)
+        gen = torch.Generator().manual_seed(5)
+        out = []
+        for _ in range(3):
+            mel = torch.randn(2, 80, 16, generator=gen)
+            audio = torch.rand(2, 1, 4096, generator=gen) * 1.6 - 0.8
+            g_loss, _, d_loss = tr.train_step(mel, audio)
+            out.append((float(g_loss), float(d_loss)
+                        if d_loss is not None else None))
+        return out
+
+    assert run() == run()
