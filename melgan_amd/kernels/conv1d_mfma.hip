@@ -1315,7 +1315,7 @@ torch::Tensor conv_transpose1d_fwd_mfma(torch::Tensor x, torch::Tensor w,
                                         int64_t act, double slope,
                                         bool act_before, int64_t To) {
   const int B = x.size(0), Cin = x.size(1), Tin = x.size(2);
-  const int Cout = w.size(1), K = w.size(2);
+  const int K = w.size(2);
   auto stream = at::hip::getCurrentHIPStream();
   // w' [Cout][Cin][k'] = w[ci][co][K-1-k']
   auto wf = w.permute({1, 0, 2}).flip(-1).contiguous();

@@ -501,7 +501,7 @@ xstage_bwd_weight_kernel(const unsigned short* __restrict__ gz,
   const int mt = blockIdx.y - g * mtiles;
   const int co_base = g * Copg + mt * 16;
   const int co_end = (g + 1) * Copg;
-  const int N = Cpg * KP;       // == 192 (eligibility: Cpg*KP == 192)
+  // N dimension = Cpg*KP == 192 (the eligibility gate guarantees it)
   constexpr int NFRAG = 3;      // per-wave 16-col fragments (static:
                                 // runtime-indexed acc goes to scratch)
 
