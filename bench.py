@@ -66,6 +66,8 @@ def main():
     ap.add_argument("--inference-rtf", dest="rtf", action="store_true",
                     help="(default) also measure inference RTF")
     args = ap.parse_args()
+    if args.steps < 1 or args.warmup < 0:
+        raise SystemExit("bench.py: --steps must be >= 1 and --warmup >= 0")
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
