@@ -148,8 +148,10 @@ def test_bench_gpus2_reexecs_torchrun(tmp_path):
     cfg = tmp_path / "cfg.yaml"
     cfg.write_text(TINY_CFG.format(d="null", ck=str(tmp_path / "ck"),
                                    lg=str(tmp_path / "lg")))
+    # NO --no-rtf: the driver's SCALE invocation runs with RTF on, so the
+    # rank-0-only RTF measurement at world>1 must be covered too
     r = _run(["bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
-              "--config", str(cfg), "--batch-size", "2", "--no-rtf"])
+              "--config", str(cfg), "--batch-size", "2"])
     assert r.returncode == 0, r.stderr[-2000:]
     recs = []
     for line in r.stdout.splitlines():  # torchrun interleaves rank output
@@ -167,6 +169,7 @@ def test_bench_gpus2_reexecs_torchrun(tmp_path):
     ov = rec["config"]["dp_overlap"]
     assert ov["g"]["buckets"] >= 1 and ov["g"]["bytes"] > 0
     assert len(ov["g"]["bucket_ms"]) == ov["g"]["buckets"]
+    assert rec["config"]["inference_rtf_22k"] > 0  # rank-0 RTF at world>1
 
 
 @pytest.mark.timeout(600)
