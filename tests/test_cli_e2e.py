@@ -138,6 +138,7 @@ def test_bench_cpu_contract(tmp_path):
     assert rec["n_gpus"] == 1 and rec["steps"] == 2
     assert rec["value"] > 0
     assert rec["config"]["parallelism"] == "dp1"
+    assert rec["config"]["inference_rtf_22k"] > 0  # RTF on by default
 
 
 @pytest.mark.timeout(600)
