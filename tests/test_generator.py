@@ -117,3 +117,18 @@ def test_cpu_singleband_forward_golden():
         y[0, 0, :8].tolist()
     assert abs(float(y.double().sum()) - 348.20916636) < 1e-2
     assert abs(float(y.mean()) - 0.13601921) < 1e-5
+
+
+@pytest.mark.parametrize("ratios", [[5, 5, 3], [5, 4, 3], [3, 3, 2]])
+def test_chunked_inference_odd_ratios(ratios):
+    """Overlap-discard must stay EXACT through odd upsample ratios
+    (k=2r, output_padding=1 convT geometry — the 24k VCTK config's
+    [5,5,3] path)."""
+    g = Generator(80, 4, 32, ratios, 3).eval()
+    torch.manual_seed(0)
+    mel = torch.randn(1, 80, 300)
+    with torch.no_grad():
+        full = g(mel)
+        chunked = g.inference_chunked(mel, chunk_frames=64, overlap=16)
+    assert chunked.shape == full.shape
+    assert torch.allclose(chunked, full, atol=1e-5)
