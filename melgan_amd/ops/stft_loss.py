@@ -84,6 +84,12 @@ class MultiResolutionSTFTLoss(nn.Module):
         super().__init__()
         if resolutions is None:
             resolutions = SUBBAND_RESOLUTIONS if subband else FULLBAND_RESOLUTIONS
+        elif isinstance(resolutions, dict):
+            # the reference family's YAML shape (train.stft_loss_params):
+            #   {fft_sizes: [...], hop_sizes: [...], win_lengths: [...]}
+            resolutions = list(zip(resolutions["fft_sizes"],
+                                   resolutions["hop_sizes"],
+                                   resolutions["win_lengths"]))
         self.losses = nn.ModuleList(
             [STFTLoss(f, h, w) for (f, h, w) in resolutions])
 

@@ -76,3 +76,38 @@ def test_fp32_under_bf16_inputs():
     y = torch.randn(1, 4096, dtype=torch.bfloat16)
     sc, mag = loss(x, y)
     assert sc.dtype == torch.float32 and mag.dtype == torch.float32
+
+
+def test_family_dict_resolutions_schema():
+    """The reference family writes train.stft_loss_params as a dict of
+    lists (fft_sizes/hop_sizes/win_lengths) — both that and the list of
+    (fft, hop, win) triples must construct the same loss."""
+    from melgan_amd.config import HParam, merge_default
+    from melgan_amd.ops.stft_loss import MultiResolutionSTFTLoss
+
+    raw = """
+train:
+  stft_loss_params:
+    fft_sizes: [1024, 2048, 512]
+    hop_sizes: [120, 240, 50]
+    win_lengths: [600, 1200, 240]
+audio: {hop_length: 256}
+model: {out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}
+"""
+    hp = merge_default(HParam(raw=raw))
+    m_dict = MultiResolutionSTFTLoss(resolutions=hp.train.stft_loss_params)
+    m_list = MultiResolutionSTFTLoss(
+        resolutions=[(1024, 120, 600), (2048, 240, 1200), (512, 50, 240)])
+    got = [(l.fft_size, l.hop_size, l.win_length) for l in m_dict.losses]
+    want = [(l.fft_size, l.hop_size, l.win_length) for l in m_list.losses]
+    assert got == want
+
+    # and the full Trainer constructs from such a config
+    from melgan_amd.engine.trainer import Trainer
+
+    hp.log.log_dir = "/tmp/famres_logs"
+    hp.log.chkpt_dir = "/tmp/famres_ck"
+    hp.backend.amp = "none"
+    hp.backend.fused_adam = False
+    tr = Trainer(hp, "famres", enable_tb=False)
+    assert len(tr.stft_full.losses) == 3
