@@ -162,6 +162,7 @@ _KNOWN_KEYS = {
     "data": {"train_dir", "val_dir"},
     "train": {"batch_size", "num_workers", "seed", "max_steps",
               "discriminator_train_start_steps", "adam", "stft_loss_params",
+              "subband_stft_loss_params",
               "lambda_adv"},
     "audio": {"n_mel_channels", "sampling_rate", "hop_length",
               "filter_length", "win_length", "segment_length", "pad_short",

@@ -76,7 +76,11 @@ class Trainer:
         self.stft_full = MultiResolutionSTFTLoss(
             resolutions=hp.train.stft_loss_params, subband=False).to(device)
         if self.multiband and hp.model.use_subband_stft_loss:
-            self.stft_sub = MultiResolutionSTFTLoss(subband=True).to(device)
+            # family configs may carry their own sub-band resolution set
+            # (train.subband_stft_loss_params, dict-of-lists or triples)
+            self.stft_sub = MultiResolutionSTFTLoss(
+                resolutions=hp.train.get("subband_stft_loss_params"),
+                subband=True).to(device)
         else:
             self.stft_sub = None
 

@@ -111,3 +111,28 @@ model: {out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2]}
     hp.backend.fused_adam = False
     tr = Trainer(hp, "famres", enable_tb=False)
     assert len(tr.stft_full.losses) == 3
+
+
+def test_family_subband_resolutions_key():
+    """train.subband_stft_loss_params (family key) overrides the
+    sub-band resolution constants; absent -> defaults."""
+    from melgan_amd.config import HParam, merge_default, unknown_keys
+    from melgan_amd.engine.trainer import Trainer
+
+    raw = """
+train:
+  subband_stft_loss_params:
+    fft_sizes: [256, 512]
+    hop_sizes: [20, 40]
+    win_lengths: [100, 200]
+audio: {hop_length: 256}
+model: {out_channels: 4, base_channels: 32, generator_ratio: [8, 4, 2],
+        use_subband_stft_loss: true}
+log: {log_dir: /tmp/famsub_logs, chkpt_dir: /tmp/famsub_ck}
+backend: {amp: none, fused_adam: false}
+"""
+    hp = merge_default(HParam(raw=raw))
+    assert "train.subband_stft_loss_params" not in unknown_keys(hp)
+    tr = Trainer(hp, "famsub", enable_tb=False)
+    got = [(l.fft_size, l.hop_size, l.win_length) for l in tr.stft_sub.losses]
+    assert got == [(256, 20, 100), (512, 40, 200)]
