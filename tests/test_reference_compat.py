@@ -99,3 +99,56 @@ def test_our_checkpoint_loads_into_reference():
     mel = torch.randn(1, 80, 12)
     with torch.no_grad():
         assert torch.allclose(ours(mel), ref(mel), atol=1e-5)
+
+
+def _reference_tower(ndf=16, n_layers=4, df=4):
+    """Family tower: torch weight_norm convs, LeakyReLU between."""
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        wn = nn.utils.weight_norm
+
+        class RefTower(nn.Module):
+            def __init__(self):
+                super().__init__()
+                layers = [wn(nn.Conv1d(1, ndf, 15))]
+                c, groups = ndf, 1
+                for _ in range(n_layers):
+                    c_out = min(c * df, 1024)
+                    groups = min(groups * df, c_out // 4)
+                    layers.append(wn(nn.Conv1d(c, c_out, 41, stride=df,
+                                               padding=20, groups=groups)))
+                    c = c_out
+                layers.append(wn(nn.Conv1d(c, c, 5, padding=2)))
+                layers.append(wn(nn.Conv1d(c, 1, 3, padding=1)))
+                self.net = nn.ModuleList(layers)
+
+            def forward(self, x):
+                feats = []
+                x = nn.functional.pad(x, (7, 7), mode="reflect")
+                for i, m in enumerate(self.net):
+                    x = m(x)
+                    if i < len(self.net) - 1:
+                        x = nn.functional.leaky_relu(x, 0.2)
+                    feats.append(x)
+                return feats
+
+        return RefTower()
+
+
+def test_reference_discriminator_tower_loads():
+    """model_d compat: a family-composition tower's weight-norm state
+    dict loads into our fused tower and produces identical features."""
+    from melgan_amd.models.discriminator import DiscriminatorTower
+
+    torch.manual_seed(0)
+    ref = _reference_tower()
+    ours = DiscriminatorTower()
+    load_flexible(ours, ref.state_dict())
+    x = torch.randn(1, 1, 2048)
+    with torch.no_grad():
+        f_ref = ref(x)
+        f_ours = ours(x)
+    assert len(f_ref) == len(f_ours)
+    for a, b in zip(f_ours, f_ref):
+        assert a.shape == b.shape
+        assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max().item()
