@@ -66,6 +66,12 @@ class Trainer:
 
             _os.environ.setdefault("HIP_LAUNCH_BLOCKING", "1")
             _os.environ.setdefault("AMD_SERIALIZE_KERNEL", "3")
+        if hp.backend.get("use_hip_kernels") is False:
+            # debugging knob: force the PyTorch composite path even on GPU
+            # (same switch the numerics tests use)
+            import os as _os
+
+            _os.environ["MELGAN_AMD_DISABLE_HIP"] = "1"
 
         self.generator = build_generator(hp).to(device)
         self.discriminator = build_discriminator(hp).to(device)

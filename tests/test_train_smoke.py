@@ -254,3 +254,16 @@ def test_seeded_run_reproducible(tiny_mb_hp, tmp_path):
         return out
 
     assert run() == run()
+
+
+def test_use_hip_kernels_false_sets_killswitch(tiny_hp, tmp_path,
+                                               monkeypatch):
+    """backend.use_hip_kernels: false must engage the composite-path
+    kill switch (MELGAN_AMD_DISABLE_HIP) the dispatch layer honours."""
+    import os
+
+    monkeypatch.delenv("MELGAN_AMD_DISABLE_HIP", raising=False)
+    tiny_hp.log.log_dir = str(tmp_path / "logs")
+    tiny_hp.backend.use_hip_kernels = False
+    Trainer(tiny_hp, "nohip", enable_tb=False)
+    assert os.environ.get("MELGAN_AMD_DISABLE_HIP") == "1"
