@@ -262,7 +262,9 @@ def test_use_hip_kernels_false_sets_killswitch(tiny_hp, tmp_path,
     kill switch (MELGAN_AMD_DISABLE_HIP) the dispatch layer honours."""
     import os
 
-    monkeypatch.delenv("MELGAN_AMD_DISABLE_HIP", raising=False)
+    # register the key with monkeypatch so the Trainer's write is
+    # restored at teardown (no leak into later tests)
+    monkeypatch.setenv("MELGAN_AMD_DISABLE_HIP", "0")
     tiny_hp.log.log_dir = str(tmp_path / "logs")
     tiny_hp.backend.use_hip_kernels = False
     Trainer(tiny_hp, "nohip", enable_tb=False)
