@@ -39,7 +39,7 @@ def bench_op(fn, reps=30, warmup=5):
 def main():
     import melgan_amd._C as C
 
-    assert torch.cuda.is_available()
+    assert torch.cuda.is_available(), "GPU micro-bench: run on an MI355X box (gpurun)"
     for tag, B, Cin, Cout, K, s, d, g, p, T in SHAPES:
         x = torch.randn(B, Cin, T, device="cuda").to(torch.bfloat16)
         w = (torch.randn(Cout, Cin // g, K, device="cuda") * 0.1) \
